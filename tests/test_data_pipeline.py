@@ -1,0 +1,93 @@
+"""Tests of the data pipeline: bridge, augment, synthetic dataset, CLIs."""
+
+import numpy as np
+import torch
+
+from waternet_amd.data.augment import PairedAugment
+from waternet_amd.data.bridge import arr2ten, ten2arr
+from waternet_amd.data.dataset import SyntheticUIEBDataset
+
+
+def test_arr2ten_hwc():
+    arr = np.arange(2 * 3 * 3, dtype=np.uint8).reshape(2, 3, 3)
+    ten = arr2ten(arr)
+    assert ten.shape == (3, 2, 3)
+    assert torch.allclose(ten[0, 0, 0], torch.tensor(0.0))
+    assert torch.allclose(ten[2, 1, 2], torch.tensor(17 / 255))
+
+
+def test_arr2ten_batchdim():
+    arr = np.zeros((4, 4, 3), dtype=np.uint8)
+    assert arr2ten(arr, add_batch_dim=True).shape == (1, 3, 4, 4)
+    batched = np.zeros((2, 4, 4, 3), dtype=np.uint8)
+    assert arr2ten(batched).shape == (2, 3, 4, 4)
+
+
+def test_ten2arr_roundtrip():
+    arr = np.random.default_rng(0).integers(
+        0, 256, size=(3, 8, 8, 3), dtype=np.uint8
+    )
+    back = ten2arr(arr2ten(arr))
+    assert back.shape == arr.shape
+    # /255 then *255 truncation: exact roundtrip
+    assert np.array_equal(back, arr)
+
+
+def test_ten2arr_clips():
+    t = torch.tensor([[[[-0.5, 0.5], [1.5, 1.0]]]])
+    arr = ten2arr(t)
+    assert arr.min() == 0 and arr.max() == 255
+
+
+def test_paired_augment_joint():
+    rng = np.random.default_rng(0)
+    aug = PairedAugment(rng=np.random.default_rng(42))
+    img = rng.integers(0, 256, size=(16, 16, 3), dtype=np.uint8)
+    # mask = image => transformed pair must stay identical
+    for _ in range(20):
+        a, b = aug(image=img, mask=img.copy())
+        assert np.array_equal(a, b)
+
+
+def test_paired_augment_actually_augments():
+    img = np.arange(16 * 16 * 3, dtype=np.uint8).reshape(16, 16, 3)
+    aug = PairedAugment(p_hflip=1.0, p_vflip=0.0, p_rot90=0.0)
+    a, _ = aug(image=img, mask=img.copy())
+    assert np.array_equal(a, img[:, ::-1])
+
+
+def test_synthetic_dataset():
+    ds = SyntheticUIEBDataset(n_images=4, im_height=32, im_width=32)
+    item = ds[0]
+    assert set(item.keys()) == {"raw", "wb", "gc", "he", "ref"}
+    for v in item.values():
+        assert v.shape == (3, 32, 32)
+        assert v.dtype == torch.float32
+        assert 0.0 <= v.min() and v.max() <= 1.0
+    # determinism
+    item2 = SyntheticUIEBDataset(n_images=4, im_height=32, im_width=32)[0]
+    assert torch.equal(item["raw"], item2["raw"])
+
+
+def test_synthetic_dataset_no_transforms():
+    ds = SyntheticUIEBDataset(n_images=2, im_height=16, im_width=16,
+                              run_transforms=False)
+    assert set(ds[0].keys()) == {"raw", "ref"}
+
+
+def test_hubconf_tuple():
+    import hubconf
+
+    preprocess, postprocess, model = hubconf.waternet(pretrained=False)
+    rgb = np.random.default_rng(1).integers(
+        0, 256, size=(32, 32, 3), dtype=np.uint8
+    )
+    rgb_ten, wb_ten, he_ten, gc_ten = preprocess(rgb)
+    for t in (rgb_ten, wb_ten, he_ten, gc_ten):
+        assert t.shape == (1, 3, 32, 32)
+    model.eval()
+    with torch.no_grad():
+        out = model(rgb_ten, wb_ten, he_ten, gc_ten)
+    arr = postprocess(out)
+    assert arr.shape == (1, 32, 32, 3)
+    assert arr.dtype == np.uint8

@@ -1,0 +1,103 @@
+"""CPU tests of model structure, forward semantics and checkpoint schema."""
+
+import torch
+import pytest
+
+from waternet_amd.models.waternet import WaterNet
+
+
+EXPECTED_KEYS = (
+    [f"cmg.conv{i}.{p}" for i in range(1, 9) for p in ("weight", "bias")]
+    + [
+        f"{b}_refiner.conv{i}.{p}"
+        for b in ("wb", "ce", "gc")
+        for i in range(1, 4)
+        for p in ("weight", "bias")
+    ]
+)
+
+EXPECTED_SHAPES = {
+    "cmg.conv1.weight": (128, 12, 7, 7),
+    "cmg.conv2.weight": (128, 128, 5, 5),
+    "cmg.conv3.weight": (128, 128, 3, 3),
+    "cmg.conv4.weight": (64, 128, 1, 1),
+    "cmg.conv5.weight": (64, 64, 7, 7),
+    "cmg.conv6.weight": (64, 64, 5, 5),
+    "cmg.conv7.weight": (64, 64, 3, 3),
+    "cmg.conv8.weight": (3, 64, 3, 3),
+    "wb_refiner.conv1.weight": (32, 6, 7, 7),
+    "wb_refiner.conv2.weight": (32, 32, 5, 5),
+    "wb_refiner.conv3.weight": (3, 32, 3, 3),
+}
+
+
+def test_forward_shape():
+    """net.py:84-90 docstring contract: (16,3,112,112) in -> same shape out."""
+    torch.manual_seed(0)
+    model = WaterNet()
+    x = torch.randn(4, 3, 64, 64)
+    out = model(x, x, x, x)
+    assert out.shape == (4, 3, 64, 64)
+
+
+def test_forward_resolution_agnostic():
+    model = WaterNet()
+    x = torch.randn(1, 3, 96, 128)
+    assert model(x, x, x, x).shape == (1, 3, 96, 128)
+
+
+def test_state_dict_schema():
+    """Checkpoint contract: 38 tensors, exact reference key names
+    (train.py:308)."""
+    model = WaterNet()
+    sd = model.state_dict()
+    assert sorted(sd.keys()) == sorted(EXPECTED_KEYS)
+    assert len(sd) == 34  # 8 cmg convs + 9 refiner convs, weight+bias each
+    total = sum(v.numel() for v in sd.values())
+    assert total == 1_090_668
+    for k, shape in EXPECTED_SHAPES.items():
+        assert tuple(sd[k].shape) == shape, k
+
+
+def test_state_dict_roundtrip(tmp_path):
+    torch.manual_seed(1)
+    m1 = WaterNet()
+    torch.save(m1.state_dict(), tmp_path / "w.pt")
+    m2 = WaterNet()
+    m2.load_state_dict(torch.load(tmp_path / "w.pt", map_location="cpu"))
+    x = torch.randn(1, 3, 32, 32)
+    assert torch.equal(m1(x, x, x, x), m2(x, x, x, x))
+
+
+def test_gated_fusion_semantics():
+    """Output = sum of refined_i * map_i with maps from sigmoid in (0,1)."""
+    torch.manual_seed(2)
+    model = WaterNet()
+    x = torch.rand(2, 3, 32, 32)
+    wb, ce, gc = torch.rand_like(x), torch.rand_like(x), torch.rand_like(x)
+    wb_cm, ce_cm, gc_cm = model.cmg(x, wb, ce, gc)
+    for cm in (wb_cm, ce_cm, gc_cm):
+        assert cm.shape == (2, 1, 32, 32)
+        assert cm.min() > 0 and cm.max() < 1
+    manual = (
+        model.wb_refiner(x, wb) * wb_cm
+        + model.ce_refiner(x, ce) * ce_cm
+        + model.gc_refiner(x, gc) * gc_cm
+    )
+    assert torch.allclose(model(x, wb, ce, gc), manual, atol=1e-6)
+
+
+def test_refiner_relu_output_nonnegative():
+    model = WaterNet()
+    x = torch.randn(1, 3, 32, 32)
+    assert model.wb_refiner(x, x).min() >= 0
+
+
+def test_backward_produces_grads():
+    model = WaterNet()
+    x = torch.rand(1, 3, 32, 32)
+    out = model(x, x, x, x)
+    out.mean().backward()
+    grads = [p.grad for p in model.parameters()]
+    assert all(g is not None for g in grads)
+    assert all(torch.isfinite(g).all() for g in grads)

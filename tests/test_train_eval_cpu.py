@@ -1,0 +1,130 @@
+"""End-to-end CPU tests: a tiny training run through the real CLIs and the
+trainer loops (BASELINE config 1 plumbing)."""
+
+import json
+import subprocess
+import sys
+from pathlib import Path
+
+import numpy as np
+import torch
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def test_trainer_one_epoch_smoke():
+    from waternet_amd.data.dataset import SyntheticUIEBDataset
+    from waternet_amd.engine.trainer import eval_one_epoch, train_one_epoch
+    from waternet_amd.models.vgg import PerceptualModel
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(0)
+    ds = SyntheticUIEBDataset(n_images=4, im_height=32, im_width=32)
+    loader = torch.utils.data.DataLoader(ds, batch_size=2)
+    model = WaterNet()
+    vgg = PerceptualModel()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    sched = torch.optim.lr_scheduler.StepLR(opt, step_size=10000, gamma=0.1)
+
+    metrics = train_one_epoch(
+        model, loader, opt, sched, vgg, torch.device("cpu"), progress=False
+    )
+    assert set(metrics) == {"mse", "ssim", "psnr", "perceptual_loss", "loss"}
+    assert all(np.isfinite(v) for v in metrics.values())
+
+    val = eval_one_epoch(model, loader, torch.device("cpu"), vgg)
+    assert set(val) == {"mse", "ssim", "psnr", "perceptual_loss"}
+
+
+def test_training_reduces_loss():
+    """A few steps of Adam on one tiny batch must reduce the composite
+    loss (sanity that gradients are correct end-to-end)."""
+    from waternet_amd.engine.losses import composite_loss
+    from waternet_amd.models.vgg import PerceptualModel
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(0)
+    model = WaterNet()
+    vgg = PerceptualModel()
+    x = torch.rand(2, 3, 32, 32)
+    ref = (x * 0.8 + 0.1).clamp(0, 1)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+
+    first = None
+    last = None
+    for _ in range(8):
+        out = model(x, x, x, x)
+        loss, _, _ = composite_loss(out, ref, vgg)
+        if first is None:
+            first = loss.item()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        last = loss.item()
+    assert last < first
+
+
+def test_train_cli_synthetic(tmp_path):
+    """Run the real train.py CLI for 1 epoch on synthetic data."""
+    out = subprocess.run(
+        [
+            sys.executable, "train.py", "--epochs", "1", "--batch-size", "2",
+            "--height", "32", "--width", "32", "--synthetic", "4",
+            "--seed", "0",
+        ],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr
+    # Find the most recent training dir and check artifacts
+    training = REPO / "training"
+    runs = sorted(
+        (p for p in training.iterdir() if p.stem.isdecimal()),
+        key=lambda p: int(p.stem),
+    )
+    savedir = runs[-1]
+    assert (savedir / "last.pt").exists()
+    assert (savedir / "metrics-train.csv").exists()
+    assert (savedir / "metrics-val.csv").exists()
+    cfg = json.loads((savedir / "config.json").read_text())
+    assert cfg["epochs"] == 1 and cfg["batch_size"] == 2
+
+    # Checkpoint is a bare state_dict loadable into a fresh model
+    from waternet_amd.models.waternet import WaterNet
+
+    sd = torch.load(savedir / "last.pt", map_location="cpu")
+    WaterNet().load_state_dict(sd)
+
+    # score.py consumes the checkpoint
+    out2 = subprocess.run(
+        [
+            sys.executable, "score.py", "--weights", str(savedir / "last.pt"),
+            "--batch-size", "2", "--height", "32", "--width", "32",
+            "--synthetic", "4", "--seed", "0",
+        ],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out2.returncode == 0, out2.stderr
+    assert "psnr" in out2.stdout
+
+
+def test_inference_cli_image(tmp_path):
+    """inference.py on a single image produces an output image."""
+    from PIL import Image
+
+    rng = np.random.default_rng(0)
+    img = rng.integers(0, 256, size=(64, 64, 3), dtype=np.uint8)
+    src = tmp_path / "test.png"
+    Image.fromarray(img).save(src)
+
+    out = subprocess.run(
+        [
+            sys.executable, "inference.py", "--source", str(src),
+            "--name", "pytest-infer",
+        ],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr
+    result = REPO / "output" / "pytest-infer" / "test.png"
+    assert result.exists()
+    with Image.open(result) as im:
+        assert im.size == (64, 64)

@@ -1,0 +1,209 @@
+"""Training entry point — CLI-compatible with the reference train.py
+(/root/reference/train.py:155-352): same flags (--epochs, --batch-size,
+--height, --width, --weights, --seed), same savedir scheme (training/<n>),
+same outputs (last.pt state_dict checkpoint each epoch, metrics-train.csv,
+metrics-val.csv, config.json).
+
+Extensions (do not break reference invocations):
+  --data-root      dataset root containing raw-890/ and reference-890/
+  --synthetic N    use N synthetic image pairs instead of UIEB on disk
+  --num-workers    DataLoader workers (reference default: 0)
+  --shuffle        shuffle the train loader (reference leaves it False)
+  --full-state     also save optimizer/scheduler/epoch sidecar for resume
+
+Multi-GPU: launch with `python -m torch.distributed.run --nproc-per-node N
+train.py ...` — one rank per GPU, RCCL all-reduce gradient sync
+(waternet_amd.parallel). Rank 0 writes checkpoints/metrics.
+"""
+
+import argparse
+import json
+import os
+from pathlib import Path
+from timeit import default_timer as timer
+
+import numpy as np
+import torch
+
+from waternet_amd.data.dataset import SyntheticUIEBDataset, UIEBDataset
+from waternet_amd.engine.trainer import (
+    TRAIN_METRICS_NAMES,
+    VAL_METRICS_NAMES,
+    eval_one_epoch,
+    train_one_epoch,
+)
+from waternet_amd.models.vgg import PerceptualModel
+from waternet_amd.models.waternet import WaterNet
+from waternet_amd.parallel import (
+    FlatBucketReducer,
+    distributed_env,
+    init_distributed,
+    shard_dataset,
+)
+
+
+def parse_args(argv=None):
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--epochs", type=int, default=400)
+    parser.add_argument("--batch-size", type=int, default=16)
+    parser.add_argument("--height", type=int, default=112)
+    parser.add_argument("--width", type=int, default=112)
+    parser.add_argument("--weights", type=str, default=None,
+                        help="(Optional) starting weights for training")
+    parser.add_argument("--seed", type=int, default=None)
+    parser.add_argument("--data-root", type=str, default="data")
+    parser.add_argument("--synthetic", type=int, default=None, metavar="N",
+                        help="use N synthetic image pairs instead of UIEB")
+    parser.add_argument("--num-workers", type=int, default=0)
+    parser.add_argument("--shuffle", action="store_true")
+    parser.add_argument("--full-state", action="store_true")
+    return parser.parse_args(argv)
+
+
+def main(argv=None):
+    start_ts = timer()
+    args = parse_args(argv)
+
+    dist_env = init_distributed()
+    rank0 = dist_env.rank == 0
+
+    projectroot = Path(__file__).parent
+    outputdir = projectroot / "training"
+    torch.manual_seed(0)  # always, as the reference (train.py:160)
+    if args.seed is not None:
+        torch.manual_seed(args.seed)
+
+    if torch.cuda.is_available():
+        torch.cuda.set_device(dist_env.local_rank)
+        device = torch.device("cuda", dist_env.local_rank)
+    else:
+        device = torch.device("cpu")
+
+    # Savedir: training/<max numbered subdir + 1>  (train.py:210-221)
+    outputdir.mkdir(exist_ok=True)
+    nums = [int(p.stem) for p in outputdir.glob("*")
+            if p.is_dir() and p.stem.isdecimal()]
+    savedir = outputdir / str(max(nums) + 1 if nums else 0)
+
+    # Dataset + split (800/90 on UIEB; proportional for synthetic)
+    if args.synthetic is not None:
+        dataset = SyntheticUIEBDataset(
+            n_images=args.synthetic, im_height=args.height, im_width=args.width
+        )
+        n_val = max(1, int(0.1 * len(dataset)))
+        split = [len(dataset) - n_val, n_val]
+    else:
+        dataset = UIEBDataset(
+            Path(args.data_root) / "raw-890",
+            Path(args.data_root) / "reference-890",
+            im_height=args.height,
+            im_width=args.width,
+        )
+        split = [800, 90]
+    train_dataset, val_dataset = torch.utils.data.random_split(dataset, split)
+    train_dataset = shard_dataset(train_dataset, dist_env)
+
+    train_loader = torch.utils.data.DataLoader(
+        train_dataset, batch_size=args.batch_size, shuffle=args.shuffle,
+        num_workers=args.num_workers, pin_memory=device.type == "cuda",
+    )
+    val_loader = torch.utils.data.DataLoader(
+        val_dataset, batch_size=args.batch_size, num_workers=args.num_workers
+    )
+
+    if rank0:
+        print(f"Using device: {device}")
+
+    model = WaterNet()
+    if args.weights is not None:
+        with open(args.weights, "rb") as f:
+            model.load_state_dict(torch.load(f, map_location="cpu"))
+    model.to(device)
+    model.train()
+
+    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
+    scheduler = torch.optim.lr_scheduler.StepLR(
+        optimizer, step_size=10000, gamma=0.1
+    )
+
+    vgg_model = PerceptualModel().to(device).eval()
+
+    reducer = None
+    if dist_env.world_size > 1:
+        # Broadcast initial params so all ranks start identical, then
+        # all-reduce gradients each step through one flat bucket.
+        reducer = FlatBucketReducer(model, dist_env)
+        reducer.broadcast_params()
+
+    saved_train = {k: [] for k in TRAIN_METRICS_NAMES}
+    saved_val = {k: [] for k in VAL_METRICS_NAMES}
+
+    for epoch in range(args.epochs):
+        train_metrics = train_one_epoch(
+            model, train_loader, optimizer, scheduler, vgg_model, device,
+            epoch_num=epoch, total_epochs=args.epochs,
+            grad_reducer=reducer, progress=rank0,
+        )
+        val_metrics = eval_one_epoch(model, val_loader, device, vgg_model)
+
+        if dist_env.world_size > 1:
+            train_metrics = dist_env.average_metrics(train_metrics)
+            val_metrics = dist_env.average_metrics(val_metrics)
+
+        if rank0:
+            print("    Train ||",
+                  "   ".join(f"{k}: {v:.03g}" for k, v in train_metrics.items()))
+            print("    Val   ||",
+                  "   ".join(f"{k}: {v:.03g}" for k, v in val_metrics.items()))
+            print()
+
+        for k, v in train_metrics.items():
+            saved_train[k].append(v)
+        for k, v in val_metrics.items():
+            saved_val[k].append(v)
+
+        if rank0:
+            savedir.mkdir(exist_ok=True)
+            torch.save(model.state_dict(), savedir / "last.pt")
+            if args.full_state:
+                torch.save(
+                    {
+                        "epoch": epoch,
+                        "optimizer": optimizer.state_dict(),
+                        "scheduler": scheduler.state_dict(),
+                    },
+                    savedir / "last-trainstate.pt",
+                )
+
+    if rank0:
+        savedir.mkdir(exist_ok=True)
+        train_arr = np.stack(
+            [np.asarray(saved_train[k]) for k in TRAIN_METRICS_NAMES], axis=1
+        )
+        val_arr = np.stack(
+            [np.asarray(saved_val[k]) for k in VAL_METRICS_NAMES], axis=1
+        )
+        np.savetxt(savedir / "metrics-train.csv", train_arr, fmt="%f",
+                   delimiter=",", comments="",
+                   header=",".join(TRAIN_METRICS_NAMES))
+        np.savetxt(savedir / "metrics-val.csv", val_arr, fmt="%f",
+                   delimiter=",", comments="",
+                   header=",".join(VAL_METRICS_NAMES))
+        with open(savedir / "config.json", "w") as f:
+            json.dump(
+                {
+                    "epochs": args.epochs,
+                    "batch_size": args.batch_size,
+                    "im_height": args.height,
+                    "im_width": args.width,
+                    "weights": args.weights,
+                },
+                f,
+                indent=4,
+            )
+        print(f"Metrics and weights saved to {savedir}")
+        print(f"Total time: {timer() - start_ts}s")
+
+
+if __name__ == "__main__":
+    main()
