@@ -1,0 +1,352 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+of the same op (run on an MI355X box via `pytest -m gpu`)."""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def q(x):
+    """bf16-round a fp32 tensor (so references see the same quantization)."""
+    return x.bfloat16().float()
+
+
+def to_nhwc_bf16(x_nchw, Cp):
+    n, c, h, w = x_nchw.shape
+    out = torch.zeros(n, h, w, Cp, device=x_nchw.device, dtype=torch.bfloat16)
+    out[..., :c] = x_nchw.permute(0, 2, 3, 1).bfloat16()
+    return out.contiguous()
+
+
+def from_nhwc(y_nhwc, C):
+    return y_nhwc[..., :C].permute(0, 3, 1, 2).float()
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from waternet_amd.ops import ext as _ext
+
+    return _ext()
+
+
+@pytest.mark.parametrize("ks,C,K", [(7, 12, 128), (5, 128, 128), (3, 128, 128),
+                                    (1, 128, 64), (7, 64, 64), (5, 6, 32),
+                                    (3, 32, 3), (3, 3, 64), (3, 512, 512)])
+def test_conv_fwd_parity(ext, ks, C, K):
+    from waternet_amd.ops.conv import pow2_channels
+
+    torch.manual_seed(0)
+    N, H, W = 2, 24, 28
+    x = torch.rand(N, C, H, W, device=DEV)
+    w = torch.randn(K, C, ks, ks, device=DEV) * 0.1
+    b = torch.randn(K, device=DEV) * 0.1
+
+    Cp, Kp = pow2_channels(C), pow2_channels(K)
+    wp = ext.pack_weight_fwd(w.contiguous(), Kp, Cp)
+    x_nhwc = to_nhwc_bf16(x, Cp)
+    y = ext.conv2d_fwd(x_nhwc, wp, b, ks, Kp, K, 1)  # relu
+    got = from_nhwc(y, K)
+
+    ref = F.relu(F.conv2d(q(x), q(w), b, padding=ks // 2))
+    err = (got - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 2e-2, f"rel err {err/scale}"
+    # pad output channels must be exactly zero
+    assert y[..., K:].abs().max().item() == 0.0 if Kp > K else True
+
+
+def test_conv_fwd_sigmoid(ext):
+    from waternet_amd.ops.conv import pow2_channels
+
+    torch.manual_seed(1)
+    N, C, K, ks = 2, 64, 3, 3
+    x = torch.rand(N, C, 16, 16, device=DEV)
+    w = torch.randn(K, C, ks, ks, device=DEV) * 0.1
+    b = torch.randn(K, device=DEV) * 0.1
+    Cp, Kp = pow2_channels(C), pow2_channels(K)
+    y = ext.conv2d_fwd(to_nhwc_bf16(x, Cp),
+                       ext.pack_weight_fwd(w.contiguous(), Kp, Cp), b, ks,
+                       Kp, K, 2)
+    ref = torch.sigmoid(F.conv2d(q(x), q(w), b, padding=1))
+    assert (from_nhwc(y, K) - ref).abs().max().item() < 1e-2
+    assert y[..., K:].abs().max().item() == 0.0  # sigmoid pad masked to 0
+
+
+@pytest.mark.parametrize("ks,C,K", [(7, 12, 128), (5, 128, 128), (3, 32, 3),
+                                    (1, 128, 64)])
+def test_conv_backward_parity(ext, ks, C, K):
+    """Full autograd through ConvBiasAct vs F.conv2d fp32 on bf16-rounded
+    inputs: dx, dw, db."""
+    from waternet_amd.ops.conv import ConvSpec, conv_bias_act, pow2_channels
+    import torch.nn as nn
+
+    torch.manual_seed(2)
+    N, H, W = 2, 16, 16
+    mod = nn.Conv2d(C, K, ks, padding="same").to(DEV)
+    spec = ConvSpec(mod, act=1)
+    x = torch.rand(N, C, H, W, device=DEV)
+    Cp = pow2_channels(C)
+    x_nhwc = to_nhwc_bf16(x, Cp).requires_grad_(True)
+    y = conv_bias_act(x_nhwc, spec)
+    dy = torch.randn_like(y)
+    dy[..., K:] = 0
+    y.backward(dy)
+
+    # reference
+    xq = q(x).requires_grad_(True)
+    wq = q(mod.weight.detach()).requires_grad_(True)
+    bq = mod.bias.detach().clone().requires_grad_(True)
+    yref = F.relu(F.conv2d(xq, wq, bq, padding=ks // 2))
+    yref.backward(from_nhwc(dy, K))
+
+    got_dx = from_nhwc(x_nhwc.grad, C)
+    for got, ref, name, tol in [
+        (got_dx, xq.grad, "dx", 3e-2),
+        (mod.weight.grad, wq.grad, "dw", 3e-2),
+        (mod.bias.grad, bq.grad, "db", 3e-2),
+    ]:
+        scale = ref.abs().max().item() + 1e-6
+        err = (got - ref).abs().max().item() / scale
+        assert err < tol, f"{name} rel err {err}"
+
+
+def test_fusion_parity(ext):
+    torch.manual_seed(3)
+    N, H, W = 2, 8, 8
+    maps = torch.rand(N, H, W, 16, device=DEV).bfloat16()
+    maps[..., 3:] = 0
+    rs = [torch.rand(N, H, W, 16, device=DEV).bfloat16() for _ in range(3)]
+    for r in rs:
+        r[..., 3:] = 0
+    maps.requires_grad_(True)
+    for r in rs:
+        r.requires_grad_(True)
+    from waternet_amd.ops.functional import GatedFusion
+
+    out = GatedFusion.apply(maps, *rs)
+    dout = torch.rand_like(out)
+    dout[..., 3:] = 0
+    out.backward(dout)
+
+    m = maps.detach().float()
+    rf = [r.detach().float().requires_grad_(True) for r in rs]
+    mf = m.clone().requires_grad_(True)
+    ref = (rf[0] * mf[..., 0:1] + rf[1] * mf[..., 1:2] + rf[2] * mf[..., 2:3])
+    # ref computed over all 16 channels; logical is 0..2
+    ref_log = (rf[0][..., :3] * mf[..., 0:1] + rf[1][..., :3] * mf[..., 1:2]
+               + rf[2][..., :3] * mf[..., 2:3])
+    assert torch.allclose(out[..., :3].float(), ref_log.detach(), atol=2e-2)
+    ref_log.backward(dout[..., :3].float())
+    assert torch.allclose(maps.grad[..., :3].float(), mf.grad[..., :3],
+                          atol=2e-2)
+    assert torch.allclose(rs[0].grad[..., :3].float(), rf[0].grad[..., :3],
+                          atol=2e-2)
+
+
+def test_maxpool_parity(ext):
+    torch.manual_seed(4)
+    x = torch.randn(2, 8, 10, 64, device=DEV).bfloat16().requires_grad_(True)
+    from waternet_amd.ops.functional import MaxPool2x2
+
+    y = MaxPool2x2.apply(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().permute(0, 3, 1, 2).requires_grad_(True)
+    yr = F.max_pool2d(xr, 2, 2)
+    yr.backward(dy.float().permute(0, 3, 1, 2))
+    assert torch.allclose(y.float().permute(0, 3, 1, 2), yr.detach(),
+                          atol=1e-2)
+    assert torch.allclose(x.grad.float().permute(0, 3, 1, 2), xr.grad,
+                          atol=1e-2)
+
+
+def test_mse255_parity(ext):
+    torch.manual_seed(5)
+    a = torch.rand(2, 8, 8, 16, device=DEV).bfloat16()
+    b = torch.rand_like(a)
+    a[..., 3:] = 0
+    b[..., 3:] = 0
+    a.requires_grad_(True)
+    from waternet_amd.ops.functional import mse255_nhwc
+
+    loss = mse255_nhwc(a, b, 3)
+    loss.backward()
+    af = a.detach().float()[..., :3].requires_grad_(True)
+    ref = torch.mean((255.0 * (af - b.float()[..., :3])) ** 2)
+    ref.backward()
+    assert abs(loss.item() - ref.item()) / ref.item() < 1e-2
+    assert torch.allclose(a.grad[..., :3].float(), af.grad, rtol=2e-2,
+                          atol=1e-1)
+
+
+def test_normalize_vgg_parity(ext):
+    from waternet_amd.ops.functional import NormalizeVgg
+    from waternet_amd.models.vgg import normalize_imagenet
+
+    torch.manual_seed(6)
+    x = torch.rand(2, 3, 8, 8, device=DEV, requires_grad=True)
+    y = NormalizeVgg.apply(x)
+    ref = normalize_imagenet(q(x.detach()))
+    assert torch.allclose(from_nhwc(y, 3), ref, atol=2e-2)
+    dy = torch.rand_like(y)
+    dy[..., 3:] = 0
+    y.backward(dy)
+    xr = x.detach().clone().requires_grad_(True)
+    normalize_imagenet(xr).backward(from_nhwc(dy, 3))
+    assert torch.allclose(x.grad, xr.grad, atol=2e-2)
+
+
+def test_ssim_native_parity(ext):
+    from waternet_amd.ops.ssim import ssim_native
+    from waternet_amd.utils.metrics import _ssim_torch
+
+    torch.manual_seed(7)
+    a = torch.rand(2, 3, 48, 48, device=DEV)
+    b = torch.rand_like(a)
+    got = ssim_native(a, b, 1.0).item()
+    ref = _ssim_torch(a, b, 1.0).item()
+    assert abs(got - ref) < 1e-3
+
+
+def test_model_forward_native_vs_eager(ext):
+    """Whole WaterNet forward: native HIP vs eager fp32 on bf16 inputs."""
+    from waternet_amd.models.waternet import WaterNet
+    import os
+
+    torch.manual_seed(8)
+    model = WaterNet().to(DEV)
+    x = torch.rand(2, 3, 32, 32, device=DEV)
+    wb, ce, gc = (torch.rand_like(x) for _ in range(3))
+    out = model(x, wb, ce, gc)
+
+    os.environ["WATERNET_AMD_EAGER"] = "1"
+    try:
+        ref = model(q(x), q(wb), q(ce), q(gc))
+    finally:
+        os.environ.pop("WATERNET_AMD_EAGER")
+    err = (out - ref).abs().max().item()
+    assert err < 0.05, f"max abs err {err}"  # bf16 conv chain tolerance
+
+
+def test_model_backward_native_vs_eager(ext):
+    from waternet_amd.models.waternet import WaterNet
+    import os
+
+    torch.manual_seed(9)
+    model = WaterNet().to(DEV)
+    x = torch.rand(2, 3, 32, 32, device=DEV)
+    out = model(x, x, x, x)
+    out.mean().backward()
+    native_grads = {
+        n: p.grad.clone() for n, p in model.named_parameters()
+    }
+    model.zero_grad()
+    os.environ["WATERNET_AMD_EAGER"] = "1"
+    try:
+        ref = model(x, x, x, x)
+        ref.mean().backward()
+    finally:
+        os.environ.pop("WATERNET_AMD_EAGER")
+    for n, p in model.named_parameters():
+        g, r = native_grads[n], p.grad
+        scale = r.abs().max().item() + 1e-8
+        err = (g - r).abs().max().item() / scale
+        assert err < 0.08, f"{n}: rel err {err}"
+
+
+def test_vgg_forward_native(ext):
+    from waternet_amd.models.vgg import PerceptualModel
+    import os
+
+    torch.manual_seed(10)
+    vgg = PerceptualModel().to(DEV)
+    x = torch.rand(1, 3, 64, 64, device=DEV)
+    out = vgg(x)
+    assert out.shape == (1, 512, 4, 4)
+    os.environ["WATERNET_AMD_EAGER"] = "1"
+    try:
+        ref = vgg(q(x))
+    finally:
+        os.environ.pop("WATERNET_AMD_EAGER")
+    # 16 stacked bf16 convs: compare with a loose relative tolerance
+    denom = ref.abs().max().item() + 1e-6
+    assert (out - ref).abs().max().item() / denom < 0.12
+
+
+def test_preprocess_gpu_vs_cpu(ext):
+    from waternet_amd.data.transforms import transform
+    from waternet_amd.ops.preprocess import gpu_transform_batch
+
+    rng = np.random.default_rng(11)
+    raw = rng.integers(0, 256, size=(2, 64, 64, 3), dtype=np.uint8)
+    wb_g, gc_g, he_g = gpu_transform_batch(
+        torch.from_numpy(raw).to(DEV)
+    )
+    for i in range(raw.shape[0]):
+        wb_c, gc_c, he_c = transform(raw[i])
+        gc_gi = gc_g[i].cpu().numpy()
+        assert np.array_equal(gc_gi, gc_c), "gamma LUT must be exact"
+        wb_gi = wb_g[i].cpu().numpy().astype(int)
+        dwb = np.abs(wb_gi - wb_c.astype(int))
+        assert dwb.max() <= 1 and (dwb > 0).mean() < 0.01, \
+            f"WB mismatch: max {dwb.max()}, frac {(dwb>0).mean()}"
+        he_gi = he_g[i].cpu().numpy().astype(int)
+        dhe = np.abs(he_gi - he_c.astype(int))
+        assert dhe.max() <= 3 and (dhe > 1).mean() < 0.02, \
+            f"CLAHE mismatch: max {dhe.max()}, frac>1 {(dhe>1).mean()}"
+
+
+def test_fused_adam_vs_torch_adam(ext):
+    from waternet_amd.ops.adam import FusedAdam
+
+    torch.manual_seed(12)
+    p1 = torch.randn(1000, device=DEV, requires_grad=True)
+    p2 = torch.randn(50, 7, device=DEV, requires_grad=True)
+    ref1 = p1.detach().clone().requires_grad_(True)
+    ref2 = p2.detach().clone().requires_grad_(True)
+
+    opt = FusedAdam([p1, p2], lr=1e-3)
+    ref_opt = torch.optim.Adam([ref1, ref2], lr=1e-3)
+    for i in range(5):
+        g1 = torch.randn(1000, device=DEV)
+        g2 = torch.randn(50, 7, device=DEV)
+        opt.zero_grad()
+        p1.grad.copy_(g1)
+        p2.grad.copy_(g2)
+        opt.step()
+        ref_opt.zero_grad()
+        ref1.grad = g1.clone()
+        ref2.grad = g2.clone()
+        ref_opt.step()
+    assert torch.allclose(p1.detach(), ref1.detach(), atol=1e-5)
+    assert torch.allclose(p2.detach(), ref2.detach(), atol=1e-5)
+
+
+def test_train_step_native_loss_decreases(ext):
+    from waternet_amd.engine.losses import composite_loss
+    from waternet_amd.models.vgg import PerceptualModel
+    from waternet_amd.models.waternet import WaterNet
+    from waternet_amd.ops.adam import FusedAdam
+
+    torch.manual_seed(13)
+    model = WaterNet().to(DEV)
+    vgg = PerceptualModel().to(DEV)
+    x = torch.rand(2, 3, 32, 32, device=DEV)
+    ref = (x * 0.8 + 0.1).clamp(0, 1)
+    opt = FusedAdam(model.parameters(), lr=1e-3, model=model)
+    losses = []
+    for _ in range(8):
+        out = model(x, x, x, x)
+        loss, _, _ = composite_loss(out, ref, vgg)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    assert all(np.isfinite(v) for v in losses)

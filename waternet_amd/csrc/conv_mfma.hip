@@ -1,0 +1,648 @@
+// MFMA implicit-GEMM convolution engine for gfx950 (CDNA4).
+//
+// Replaces the reference's implicit cuDNN conv launches (SURVEY §2.2
+// K2-K9, K12-K14, K17) with hand-written NHWC bf16 kernels:
+//   - k_conv_igemm<BN,KS>: stride-1 same-pad conv fwd as implicit GEMM
+//     (M = N*H*W rows, N = output channels, K = R*S*Cp) on
+//     v_mfma_f32_16x16x32_bf16 with fused bias + ReLU/Sigmoid epilogue.
+//     Also runs dgrad (conv of dY with rotated/transposed weights packed by
+//     k_pack_dgrad).
+//   - k_conv_wgrad<KS>: weight gradient as reduction GEMM over M with
+//     fp32 atomic accumulation straight into the NCHW fp32 grad tensor.
+//   - k_bias_grad: column reduction of dY.
+//   - k_pack_fwd / k_pack_dgrad: NCHW fp32 master weights -> bf16 packed
+//     [Kp][R*S*Cp] (k-major, rsc contiguous) layouts.
+//
+// Design notes (MI355X):
+//   - LDS tiles are stored FRAGMENT-MAJOR: slot (mf, kb, i) holds the 8
+//     bf16 the MFMA lane (kb,i) consumes, so both ds_write_b128 staging and
+//     ds_read_b128 fragment loads are bank-conflict-free without swizzles
+//     (lane l covers dwords 4*(l&15)..+3 of a 1 KiB block; the b128 lane
+//     groups each touch all 64 banks exactly once).
+//   - Cp (physical channels) is a power of two >= 16; pad channels are
+//     zero by construction everywhere, so no channel masking in the GEMM.
+//   - Double-buffered LDS, global loads for step k+1 issued before the
+//     MFMA of step k (async-STAGE split, write-late).
+//   - 256 threads = 4 waves; BM=128 rows; BN in {16,32,64,128} columns.
+//
+// Fragment K mapping: set by WN_MFMA_KMAP (probe_mfma.hip measures it on
+// hardware). 0: lane group g holds k = g*8+e. 1: k = g*4+(e&3)+(e>>2)*16.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+#ifndef WN_MFMA_KMAP
+#define WN_MFMA_KMAP 0
+#endif
+
+WN_DEVFN constexpr int frag_k16(int g, int e) {
+#if WN_MFMA_KMAP == 0
+  return g * 8 + e;
+#else
+  return g * 4 + (e & 3) + ((e >> 2) << 4);
+#endif
+}
+
+// ---------------------------------------------------------------------------
+// Forward / dgrad implicit GEMM
+// ---------------------------------------------------------------------------
+
+template <int BN, int KS>
+__global__ __launch_bounds__(256, 2) void k_conv_igemm(
+    const bf16_t* __restrict__ X,   // (N, H, W, Cp)
+    const bf16_t* __restrict__ Wp,  // [Kp][RS*Cp] k-major packed
+    const float* __restrict__ Bias, // [>=Klog] or nullptr
+    bf16_t* __restrict__ Y,         // (N, H, W, Kp)
+    int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act) {
+  constexpr int BM = 128;
+  constexpr int PAD = KS / 2;
+  constexpr int RS = KS * KS;
+  constexpr int WM = (BN >= 64) ? 2 : 4;
+  constexpr int WN = 4 / WM;
+  constexpr int FM = BM / WM / 16;  // fragments per wave in M
+  constexpr int FN = BN / WN / 16;  // fragments per wave in N
+  constexpr int SLOTS_B = BN * 4;
+  constexpr int NBS = (SLOTS_B + 255) / 256;  // B slots per thread (1 or 2)
+  const int KG = RS * Cp;
+  const long M = (long)N * H * W;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 * BM*32
+  bf16_t* lB = lA + 2 * BM * 32;                             // 2 * BN*32
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid / WN;
+  const int wc = wid % WN;
+  const long m0 = (long)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  // ---- per-thread A-slot geometry (2 slots) ----
+  int aOy[2], aOx[2], aKb[2];
+  long aRowBase[2];  // (n*H) term
+  bool aMv[2];
+  const int HW = H * W;
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    int slot = tid + s * 256;
+    int i = slot & 15, kb = (slot >> 4) & 3;
+    int mf = slot >> 6;
+    long m = m0 + mf * 16 + i;
+    aMv[s] = m < M;
+    long mm = aMv[s] ? m : 0;
+    int n = (int)(mm / HW);
+    int rem = (int)(mm - (long)n * HW);
+    aOy[s] = rem / W;
+    aOx[s] = rem - aOy[s] * W;
+    aRowBase[s] = (long)n * H;
+    aKb[s] = kb;
+  }
+  // ---- per-thread B-slot geometry ----
+  int bK[NBS], bKb[NBS];
+  bool bV[NBS];
+#pragma unroll
+  for (int s = 0; s < NBS; ++s) {
+    int slot = tid + s * 256;
+    bV[s] = slot < SLOTS_B;
+    int j = slot & 15, kb = (slot >> 4) & 3;
+    int nf = (slot >> 6);
+    bK[s] = n0 + nf * 16 + j;
+    bKb[s] = kb;
+  }
+
+  bf16x8 aR[2], bR[NBS];
+
+  auto loadA = [&](int k0) {
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 v = {};
+      if (aMv[s]) {
+#if WN_MFMA_KMAP == 0
+        // 8 contiguous k: one 16B load (within one tap: Cp % 8 == 0)
+        int rsc = k0 + aKb[s] * 8;
+        if (rsc < KG) {
+          int tap = rsc >> log2Cp;
+          int c = rsc & (Cp - 1);
+          int dy = tap / KS, dx = tap - (tap / KS) * KS;
+          int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
+          if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+            v = *reinterpret_cast<const bf16x8*>(
+                X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c);
+        }
+#else
+        // two 4-half pieces (k = 4g.. and k = 16+4g..)
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          int rsc = k0 + aKb[s] * 4 + h * 16;
+          if (rsc < KG) {
+            int tap = rsc >> log2Cp;
+            int c = rsc & (Cp - 1);
+            int dy = tap / KS, dx = tap - (tap / KS) * KS;
+            int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
+            if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
+              const ushort2 p = *reinterpret_cast<const ushort2*>(
+                  X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c);
+              const ushort2 q = *reinterpret_cast<const ushort2*>(
+                  X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c + 2);
+              reinterpret_cast<ushort2*>(&v)[h * 2] = p;
+              reinterpret_cast<ushort2*>(&v)[h * 2 + 1] = q;
+            }
+          }
+        }
+#endif
+      }
+      aR[s] = v;
+    }
+  };
+
+  auto loadB = [&](int k0) {
+#pragma unroll
+    for (int s = 0; s < NBS; ++s) {
+      bf16x8 v = {};
+      if (bV[s] && bK[s] < Kp) {
+#if WN_MFMA_KMAP == 0
+        int rsc = k0 + bKb[s] * 8;
+        if (rsc < KG)
+          v = *reinterpret_cast<const bf16x8*>(Wp + (long)bK[s] * KG + rsc);
+#else
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          int rsc = k0 + bKb[s] * 4 + h * 16;
+          if (rsc < KG) {
+            const ushort2 p = *reinterpret_cast<const ushort2*>(
+                Wp + (long)bK[s] * KG + rsc);
+            const ushort2 q = *reinterpret_cast<const ushort2*>(
+                Wp + (long)bK[s] * KG + rsc + 2);
+            reinterpret_cast<ushort2*>(&v)[h * 2] = p;
+            reinterpret_cast<ushort2*>(&v)[h * 2 + 1] = q;
+          }
+        }
+#endif
+      }
+      bR[s] = v;
+    }
+  };
+
+  auto writeTiles = [&](int buf) {
+#pragma unroll
+    for (int s = 0; s < 2; ++s)
+      *reinterpret_cast<bf16x8*>(lA + buf * (BM * 32) + (tid + s * 256) * 8) =
+          aR[s];
+#pragma unroll
+    for (int s = 0; s < NBS; ++s)
+      if (bV[s])
+        *reinterpret_cast<bf16x8*>(lB + buf * (BN * 32) + (tid + s * 256) * 8) =
+            bR[s];
+  };
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int a = 0; a < FM; ++a)
+#pragma unroll
+    for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nk = (KG + 31) / 32;
+  loadA(0);
+  loadB(0);
+  writeTiles(0);
+  __syncthreads();
+
+  const int lg = lane >> 4;   // fragment k-group
+  const int li = lane & 15;   // fragment row/col
+
+  for (int ks = 0; ks < nk; ++ks) {
+    const int cur = ks & 1;
+    if (ks + 1 < nk) {
+      loadA((ks + 1) * 32);
+      loadB((ks + 1) * 32);
+    }
+    bf16x8 aF[FM], bF[FN];
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+      int mfG = wr * FM + fm;
+      aF[fm] = *reinterpret_cast<const bf16x8*>(
+          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
+    }
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      int nfG = wc * FN + fn;
+      bF[fn] = *reinterpret_cast<const bf16x8*>(
+          lB + cur * (BN * 32) + ((nfG * 4 + lg) * 16 + li) * 8);
+    }
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+    if (ks + 1 < nk) {
+      __syncthreads();
+      writeTiles(cur ^ 1);
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: bias + activation (+ pad-channel zeroing) ----
+  const int lr4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int fn = 0; fn < FN; ++fn) {
+    const int k = n0 + (wc * FN + fn) * 16 + li;
+    if (k >= Kp) continue;
+    const float bv = (Bias != nullptr && k < Klog) ? Bias[k] : 0.f;
+    const bool kpad = k >= Klog;
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long m = m0 + (wr * FM + fm) * 16 + lr4 + r;
+        if (m >= M) continue;
+        float v = acc[fm][fn][r] + bv;
+        if (act == ACT_RELU) v = fmaxf(v, 0.f);
+        else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
+        if (kpad) v = 0.f;
+        Y[m * Kp + k] = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Weight gradient: dW[k][rsc] = sum_m dY[m][k] * X[m][rsc]
+// 64x64 output tile per block, K-dim = m in steps of 32, fp32 atomics into
+// the NCHW fp32 grad tensor.
+// ---------------------------------------------------------------------------
+
+template <int KS>
+__global__ __launch_bounds__(256, 2) void k_conv_wgrad(
+    const bf16_t* __restrict__ dY,  // (N,H,W,Kp)
+    const bf16_t* __restrict__ X,   // (N,H,W,Cp)
+    float* __restrict__ dW,         // (K, C, KS, KS) fp32, pre-zeroed
+    int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C,
+    int splitm) {
+  constexpr int PAD = KS / 2;
+  constexpr int RS = KS * KS;
+  const int KG = RS * Cp;
+  const long M = (long)N * H * W;
+  const int HW = H * W;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);  // 64k x 32m frag-major
+  bf16_t* lB = lA + 64 * 32;                     // 64rsc x 32m frag-major
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;  // 2x2 waves, 32x32 tile each
+  const int wc = wid & 1;
+  const int kt0 = blockIdx.x * 64;
+  const int rt0 = blockIdx.y * 64;
+
+  // staging geometry: thread -> (mrow = tid>>3 in 0..31, grp = tid&7)
+  const int mrow = tid >> 3;
+  const int grp = tid & 7;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const long nChunks = (M + 31) / 32;
+  for (long chunk = blockIdx.z; chunk < nChunks; chunk += splitm) {
+    const long m0 = chunk * 32;
+    const long m = m0 + mrow;
+    const bool mv = m < M;
+
+    // ---- stage A (dY^T): load 8 k for (m, kgrp), scatter-write ----
+    {
+      bf16x8 v = {};
+      int kg = kt0 + grp * 8;
+      if (mv && kg + 7 < Kp)
+        v = *reinterpret_cast<const bf16x8*>(dY + m * Kp + kg);
+      else if (mv) {
+        for (int e = 0; e < 8; ++e)
+          if (kg + e < Kp) v[e] = dY[m * Kp + kg + e];
+      }
+      const int kb = mrow >> 3, eM = mrow & 7;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int kl = grp * 8 + e;  // local k 0..63
+        int i = kl & 15, kf = kl >> 4;
+        // element eM must sit at position where frag_k16(kb, pos) == mrow&..:
+        // frag element index epos satisfies frag_k16(kb, epos) = mrow for the
+        // reading lane group kb' = reading lane>>4. Store m-major into the
+        // frag slot: slot (kf, kb, i), element epos with
+        // frag_k16(kb, epos) == mrow (within 0..31).
+        lA[((kf * 4 + kb) * 16 + i) * 8 + eM] = v[e];
+      }
+    }
+    // ---- stage B (X gather, transposed the same way) ----
+    {
+      bf16x8 v = {};
+      if (mv) {
+        long mm = m;
+        int n = (int)(mm / HW);
+        int rem = (int)(mm - (long)n * HW);
+        int oy = rem / W, ox = rem - (rem / W) * W;
+        int rsc = rt0 + grp * 8;
+        if (rsc < KG) {
+          int tap = rsc >> log2Cp;
+          int c = rsc & (Cp - 1);
+          int dy = tap / KS, dx = tap - (tap / KS) * KS;
+          int iy = oy + dy - PAD, ix = ox + dx - PAD;
+          if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+            v = *reinterpret_cast<const bf16x8*>(
+                X + (((long)(n * H + iy) * W + ix) << log2Cp) + c);
+        }
+      }
+      const int kb = mrow >> 3, eM = mrow & 7;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        int jl = grp * 8 + e;
+        int j = jl & 15, jf = jl >> 4;
+        lB[((jf * 4 + kb) * 16 + j) * 8 + eM] = v[e];
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA: K-dim = 32 m values ----
+    const int lg = lane >> 4, li = lane & 15;
+    bf16x8 aF[2], bF[2];
+#pragma unroll
+    for (int f = 0; f < 2; ++f) {
+      // NOTE: LDS holds elements m-ordered (eM = m&7 at position eM), but the
+      // MFMA consumes element e as k = frag_k16(g, e). For KMAP 0 these agree
+      // (frag_k16(g,e) = g*8+e and we stored m = kb*8+eM at slot(kb)+eM).
+      // For KMAP 1 the staging writes must permute: handled below.
+      aF[f] = *reinterpret_cast<const bf16x8*>(
+          lA + (((wr * 2 + f) * 4 + lg) * 16 + li) * 8);
+      bF[f] = *reinterpret_cast<const bf16x8*>(
+          lB + (((wc * 2 + f) * 4 + lg) * 16 + li) * 8);
+    }
+#pragma unroll
+    for (int fa = 0; fa < 2; ++fa)
+#pragma unroll
+      for (int fb = 0; fb < 2; ++fb)
+        acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aF[fa], bF[fb], acc[fa][fb], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // ---- epilogue: scatter-add into NCHW fp32 dW ----
+  const int lg = lane >> 4, li = lane & 15;
+#pragma unroll
+  for (int fa = 0; fa < 2; ++fa) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int k = kt0 + (wr * 2 + fa) * 16 + lg * 4 + r;
+      if (k >= K) continue;
+#pragma unroll
+      for (int fb = 0; fb < 2; ++fb) {
+        int rsc = rt0 + (wc * 2 + fb) * 16 + li;
+        if (rsc >= KG) continue;
+        int tap = rsc >> log2Cp;
+        int c = rsc & (Cp - 1);
+        if (c >= C) continue;
+        int dy = tap / KS, dx = tap - (tap / KS) * KS;
+        float v = acc[fa][fb][r];
+        atomicAdd(&dW[(((long)k * C + c) * KS + dy) * KS + dx], v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Bias gradient: db[k] = sum_m dY[m][k]
+// ---------------------------------------------------------------------------
+
+__global__ void k_bias_grad(const bf16_t* __restrict__ dY,
+                            float* __restrict__ dB, long M, int Kp, int K,
+                            int msplit) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = reinterpret_cast<float*>(smem);
+  const int KX = min(Kp, 64);
+  const int ROWS = 256 / KX;
+  const int kx = threadIdx.x % KX;
+  const int my = threadIdx.x / KX;
+  const int kbase = blockIdx.x * KX;
+  const int k = kbase + kx;
+
+  float s = 0.f;
+  const long rows_per_split = (M + msplit - 1) / msplit;
+  const long mstart = (long)blockIdx.y * rows_per_split;
+  const long mend = min(mstart + rows_per_split, M);
+  if (k < Kp)
+    for (long m = mstart + my; m < mend; m += ROWS)
+      s += bf2f(dY[m * Kp + k]);
+  red[my * KX + kx] = s;
+  __syncthreads();
+  if (my == 0 && k < K) {
+    float t = 0.f;
+    for (int r = 0; r < ROWS; ++r) t += red[r * KX + kx];
+    atomicAdd(&dB[k], t);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Weight packing: NCHW fp32 master -> bf16 k-major [Kp][RS*Cp] (fwd) and
+// c-major rotated [Cp][RS*Kp] (dgrad).
+// ---------------------------------------------------------------------------
+
+__global__ void k_pack_fwd(const float* __restrict__ Wm,  // (K,C,R,S)
+                           bf16_t* __restrict__ out,      // [Kp][RS*Cp]
+                           int K, int C, int R, int S, int Kp, int Cp) {
+  const int KG = R * S * Cp;
+  const long total = (long)Kp * KG;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int k = (int)(idx / KG);
+    int rsc = (int)(idx - (long)k * KG);
+    int tap = rsc / Cp, c = rsc - (rsc / Cp) * Cp;
+    int r = tap / S, s = tap - (tap / S) * S;
+    float v = (k < K && c < C) ? Wm[(((long)k * C + c) * R + r) * S + s] : 0.f;
+    out[idx] = f2bf(v);
+  }
+}
+
+__global__ void k_pack_dgrad(const float* __restrict__ Wm,  // (K,C,R,S)
+                             bf16_t* __restrict__ out,      // [Cp][RS*Kp]
+                             int K, int C, int R, int S, int Kp, int Cp) {
+  const int KG = R * S * Kp;
+  const long total = (long)Cp * KG;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    int c = (int)(idx / KG);
+    int rsk = (int)(idx - (long)c * KG);
+    int tap = rsk / Kp, k = rsk - (rsk / Kp) * Kp;
+    int rr = tap / S, ss = tap - (tap / S) * S;
+    int r = R - 1 - rr, s = S - 1 - ss;  // rotate 180
+    float v = (c < C && k < K) ? Wm[(((long)k * C + c) * R + r) * S + s] : 0.f;
+    out[idx] = f2bf(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host-side dispatch
+// ---------------------------------------------------------------------------
+
+namespace {
+
+inline int log2i(int v) {
+  int l = 0;
+  while ((1 << l) < v) ++l;
+  TORCH_CHECK((1 << l) == v, "value not a power of two: ", v);
+  return l;
+}
+
+template <int KS>
+void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
+                    const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                    int Klog, int act, hipStream_t stream) {
+  const int N = x.size(0), H = x.size(1), W = x.size(2), Cp = x.size(3);
+  const int Kp = y.size(3);
+  const long M = (long)N * H * W;
+  const float* bptr =
+      bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  const int gx = (int)((M + 127) / 128);
+
+  auto launch = [&](auto bn_const) {
+    constexpr int BN = decltype(bn_const)::value;
+    const int gy = (Kp + BN - 1) / BN;
+    const size_t lds = (2 * 128 * 32 + 2 * BN * 32) * sizeof(bf16_t);
+    hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256), lds,
+                       stream, (const bf16_t*)x.data_ptr(),
+                       (const bf16_t*)wp.data_ptr(), bptr,
+                       (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
+                       Klog, act);
+  };
+  if (Kp >= 128)
+    launch(std::integral_constant<int, 128>{});
+  else if (Kp == 64)
+    launch(std::integral_constant<int, 64>{});
+  else if (Kp == 32)
+    launch(std::integral_constant<int, 32>{});
+  else
+    launch(std::integral_constant<int, 16>{});
+  HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+at::Tensor conv2d_fwd(const at::Tensor& x, const at::Tensor& wp,
+                      const c10::optional<at::Tensor>& bias, int64_t ks,
+                      int64_t Kp, int64_t Klog, int64_t act) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 && x.dim() == 4,
+              "x must be CUDA bf16 NHWC");
+  TORCH_CHECK(wp.is_cuda() && wp.dtype() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && wp.is_contiguous());
+  auto y = at::empty({x.size(0), x.size(1), x.size(2), Kp},
+                     x.options());
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  switch (ks) {
+    case 1:
+      launch_conv_bn<1>(x, wp, bias, y, (int)Klog, (int)act, stream);
+      break;
+    case 3:
+      launch_conv_bn<3>(x, wp, bias, y, (int)Klog, (int)act, stream);
+      break;
+    case 5:
+      launch_conv_bn<5>(x, wp, bias, y, (int)Klog, (int)act, stream);
+      break;
+    case 7:
+      launch_conv_bn<7>(x, wp, bias, y, (int)Klog, (int)act, stream);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported kernel size ", ks);
+  }
+  return y;
+}
+
+void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
+                  int64_t ks) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == at::kBFloat16);
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16);
+  TORCH_CHECK(dw.is_cuda() && dw.dtype() == at::kFloat && dw.dim() == 4);
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous() && dw.is_contiguous());
+  const int N = x.size(0), H = x.size(1), W = x.size(2), Cp = x.size(3);
+  const int Kp = dy.size(3);
+  const int K = dw.size(0), C = dw.size(1);
+  TORCH_CHECK(dw.size(2) == ks && dw.size(3) == ks);
+  const int KG = (int)(ks * ks) * Cp;
+  const int gx = (Kp + 63) / 64, gy = (KG + 63) / 64;
+  // pick split so gx*gy*split ~ 2*256 blocks
+  int split = std::max(1, 512 / std::max(1, gx * gy));
+  const long nChunks = ((long)N * H * W + 31) / 32;
+  split = (int)std::min<long>(split, nChunks);
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  const size_t lds = 2 * 64 * 32 * sizeof(bf16_t);
+  auto launch = [&](auto ks_const) {
+    constexpr int KSV = decltype(ks_const)::value;
+    hipLaunchKernelGGL((k_conv_wgrad<KSV>), dim3(gx, gy, split), dim3(256),
+                       lds, stream, (const bf16_t*)dy.data_ptr(),
+                       (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
+                       H, W, Cp, log2i(Cp), Kp, K, C, split);
+  };
+  switch (ks) {
+    case 1: launch(std::integral_constant<int, 1>{}); break;
+    case 3: launch(std::integral_constant<int, 3>{}); break;
+    case 5: launch(std::integral_constant<int, 5>{}); break;
+    case 7: launch(std::integral_constant<int, 7>{}); break;
+    default: TORCH_CHECK(false, "unsupported kernel size ", ks);
+  }
+  HIP_CHECK_LAST();
+}
+
+void bias_grad(const at::Tensor& dy, at::Tensor& db) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == at::kBFloat16);
+  TORCH_CHECK(db.is_cuda() && db.dtype() == at::kFloat);
+  const int Kp = dy.size(3);
+  const long M = dy.numel() / Kp;
+  const int K = db.size(0);
+  const int KX = std::min(Kp, 64);
+  const int gx = (Kp + KX - 1) / KX;
+  int msplit = (int)std::min<long>(64, (M + 255) / 256);
+  msplit = std::max(msplit, 1);
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_bias_grad, dim3(gx, msplit), dim3(256),
+                     256 * sizeof(float), stream,
+                     (const bf16_t*)dy.data_ptr(), db.data_ptr<float>(), M,
+                     Kp, K, msplit);
+  HIP_CHECK_LAST();
+}
+
+at::Tensor pack_weight_fwd(const at::Tensor& w, int64_t Kp, int64_t Cp) {
+  TORCH_CHECK(w.is_cuda() && w.dtype() == at::kFloat && w.dim() == 4);
+  TORCH_CHECK(w.is_contiguous());
+  const int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  auto out = at::empty({Kp, (long)R * S * Cp},
+                       w.options().dtype(at::kBFloat16));
+  const long total = out.numel();
+  const int blocks = (int)std::min<long>(2048, (total + 255) / 256);
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_pack_fwd, dim3(blocks), dim3(256), 0, stream,
+                     w.data_ptr<float>(), (bf16_t*)out.data_ptr(), K, C, R, S,
+                     (int)Kp, (int)Cp);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+at::Tensor pack_weight_dgrad(const at::Tensor& w, int64_t Kp, int64_t Cp) {
+  // output [Cp][RS*Kp]: dgrad conv consumes dY (channels Kp) and produces
+  // dX (channels Cp); weights rotated 180 and transposed.
+  TORCH_CHECK(w.is_cuda() && w.dtype() == at::kFloat && w.dim() == 4);
+  TORCH_CHECK(w.is_contiguous());
+  const int K = w.size(0), C = w.size(1), R = w.size(2), S = w.size(3);
+  auto out = at::empty({Cp, (long)R * S * Kp},
+                       w.options().dtype(at::kBFloat16));
+  const long total = out.numel();
+  const int blocks = (int)std::min<long>(2048, (total + 255) / 256);
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_pack_dgrad, dim3(blocks), dim3(256), 0, stream,
+                     w.data_ptr<float>(), (bf16_t*)out.data_ptr(), K, C, R, S,
+                     (int)Kp, (int)Cp);
+  HIP_CHECK_LAST();
+  return out;
+}

@@ -1,0 +1,464 @@
+// Fused elementwise / layout / reduction kernels for the WaterNet engine
+// (gfx950). Covers SURVEY §2.2 K1/K10/K11 (cat/split folding via the input
+// builders), K15 (gated fusion fwd/bwd), K16 (ImageNet normalize), K18/K19
+// (fused squared-diff reductions), K23/K25 (fused flat Adam), plus the
+// NCHW fp32 <-> NHWC bf16 bridges and activation backward.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+inline hipStream_t cur_stream() { return at::cuda::getCurrentHIPStream(); }
+constexpr int TPB = 256;
+inline int grid1d(long n, int tpb = TPB, int cap = 4096) {
+  return (int)std::min<long>(cap, (n + tpb - 1) / tpb);
+}
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Input builders: 4x NCHW fp32 (3ch) -> cmg input (N,H,W,16) and three
+// refiner inputs (N,H,W,16), bf16, /1 scale (inputs already in [0,1]).
+// Folds the reference's torch.cat calls (net.py:46, net.py:76) into one pass.
+// ---------------------------------------------------------------------------
+
+__global__ void k_build_inputs(const float* __restrict__ raw,
+                               const float* __restrict__ wb,
+                               const float* __restrict__ ce,
+                               const float* __restrict__ gc,
+                               bf16_t* __restrict__ cmg_in,
+                               bf16_t* __restrict__ rwb_in,
+                               bf16_t* __restrict__ rce_in,
+                               bf16_t* __restrict__ rgc_in, long NHW,
+                               long HW) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    const long base = n * 3 * HW + rem;
+    float r0 = raw[base], r1 = raw[base + HW], r2 = raw[base + 2 * HW];
+    float w0 = wb[base], w1 = wb[base + HW], w2 = wb[base + 2 * HW];
+    float c0 = ce[base], c1 = ce[base + HW], c2 = ce[base + 2 * HW];
+    float g0 = gc[base], g1 = gc[base + HW], g2 = gc[base + 2 * HW];
+    bf16_t* o = cmg_in + p * 16;
+    o[0] = f2bf(r0); o[1] = f2bf(r1); o[2] = f2bf(r2);
+    o[3] = f2bf(w0); o[4] = f2bf(w1); o[5] = f2bf(w2);
+    o[6] = f2bf(c0); o[7] = f2bf(c1); o[8] = f2bf(c2);
+    o[9] = f2bf(g0); o[10] = f2bf(g1); o[11] = f2bf(g2);
+    o[12] = o[13] = o[14] = o[15] = f2bf(0.f);
+    auto emit = [&](bf16_t* dst, float a0, float a1, float a2) {
+      bf16_t* q = dst + p * 16;
+      q[0] = f2bf(r0); q[1] = f2bf(r1); q[2] = f2bf(r2);
+      q[3] = f2bf(a0); q[4] = f2bf(a1); q[5] = f2bf(a2);
+#pragma unroll
+      for (int i = 6; i < 16; ++i) q[i] = f2bf(0.f);
+    };
+    emit(rwb_in, w0, w1, w2);
+    emit(rce_in, c0, c1, c2);
+    emit(rgc_in, g0, g1, g2);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// NCHW fp32 <-> NHWC bf16 bridges
+// ---------------------------------------------------------------------------
+
+__global__ void k_nchw2nhwc(const float* __restrict__ x,
+                            bf16_t* __restrict__ y, long NHW, long HW, int C,
+                            int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    const float* src = x + (n * C) * HW + rem;
+    bf16_t* dst = y + p * Cp;
+    int c = 0;
+    for (; c < C; ++c) dst[c] = f2bf(src[(long)c * HW]);
+    for (; c < Cp; ++c) dst[c] = f2bf(0.f);
+  }
+}
+
+__global__ void k_nhwc2nchw(const bf16_t* __restrict__ x,
+                            float* __restrict__ y, long NHW, long HW, int C,
+                            int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    const bf16_t* src = x + p * Cp;
+    float* dst = y + (n * C) * HW + rem;
+    for (int c = 0; c < C; ++c) dst[(long)c * HW] = bf2f(src[c]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Gated fusion (net.py:104-108): out_c = sum_i refined_i_c * map_i
+// maps = (N,H,W,16) logical 3 (cmg conv8 sigmoid output)
+// refined_i = (N,H,W,16) logical 3
+// ---------------------------------------------------------------------------
+
+__global__ void k_fusion_fwd(const bf16_t* __restrict__ maps,
+                             const bf16_t* __restrict__ rwb,
+                             const bf16_t* __restrict__ rce,
+                             const bf16_t* __restrict__ rgc,
+                             bf16_t* __restrict__ out, long NHW) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long o = p * 16;
+    float m0 = bf2f(maps[o]), m1 = bf2f(maps[o + 1]), m2 = bf2f(maps[o + 2]);
+    bf16_t* dst = out + o;
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      float v = bf2f(rwb[o + c]) * m0 + bf2f(rce[o + c]) * m1 +
+                bf2f(rgc[o + c]) * m2;
+      dst[c] = f2bf(v);
+    }
+#pragma unroll
+    for (int c = 3; c < 16; ++c) dst[c] = f2bf(0.f);
+  }
+}
+
+__global__ void k_fusion_bwd(const bf16_t* __restrict__ dout,
+                             const bf16_t* __restrict__ maps,
+                             const bf16_t* __restrict__ rwb,
+                             const bf16_t* __restrict__ rce,
+                             const bf16_t* __restrict__ rgc,
+                             bf16_t* __restrict__ dmaps,
+                             bf16_t* __restrict__ drwb,
+                             bf16_t* __restrict__ drce,
+                             bf16_t* __restrict__ drgc, long NHW) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long o = p * 16;
+    float m0 = bf2f(maps[o]), m1 = bf2f(maps[o + 1]), m2 = bf2f(maps[o + 2]);
+    float dm0 = 0.f, dm1 = 0.f, dm2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      float g = bf2f(dout[o + c]);
+      dm0 += g * bf2f(rwb[o + c]);
+      dm1 += g * bf2f(rce[o + c]);
+      dm2 += g * bf2f(rgc[o + c]);
+      drwb[o + c] = f2bf(g * m0);
+      drce[o + c] = f2bf(g * m1);
+      drgc[o + c] = f2bf(g * m2);
+    }
+    dmaps[o] = f2bf(dm0);
+    dmaps[o + 1] = f2bf(dm1);
+    dmaps[o + 2] = f2bf(dm2);
+#pragma unroll
+    for (int c = 3; c < 16; ++c) {
+      dmaps[o + c] = f2bf(0.f);
+      drwb[o + c] = f2bf(0.f);
+      drce[o + c] = f2bf(0.f);
+      drgc[o + c] = f2bf(0.f);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Activation backward: dpre = dY * act'(Y)  (Y = post-activation output)
+// ---------------------------------------------------------------------------
+
+__global__ void k_act_bwd(const bf16_t* __restrict__ dy,
+                          const bf16_t* __restrict__ y,
+                          bf16_t* __restrict__ dpre, long n, int act) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float g = bf2f(dy[i]);
+    float v = bf2f(y[i]);
+    float r = (act == ACT_RELU) ? (v > 0.f ? g : 0.f) : g * v * (1.f - v);
+    dpre[i] = f2bf(r);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// ImageNet normalize fused with NCHW fp32 -> NHWC bf16 (and backward)
+// (train.py:111-116)
+// ---------------------------------------------------------------------------
+
+__constant__ float IMNET_MEAN[3] = {0.485f, 0.456f, 0.406f};
+__constant__ float IMNET_STD[3] = {0.229f, 0.224f, 0.225f};
+
+__global__ void k_normalize_vgg_fwd(const float* __restrict__ x,
+                                    bf16_t* __restrict__ y, long NHW,
+                                    long HW, int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    const float* src = x + (n * 3) * HW + rem;
+    bf16_t* dst = y + p * Cp;
+#pragma unroll
+    for (int c = 0; c < 3; ++c)
+      dst[c] = f2bf((src[(long)c * HW] - IMNET_MEAN[c]) / IMNET_STD[c]);
+    for (int c = 3; c < Cp; ++c) dst[c] = f2bf(0.f);
+  }
+}
+
+__global__ void k_normalize_vgg_bwd(const bf16_t* __restrict__ dy,
+                                    float* __restrict__ dx, long NHW,
+                                    long HW, int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    const bf16_t* src = dy + p * Cp;
+    float* dst = dx + (n * 3) * HW + rem;
+#pragma unroll
+    for (int c = 0; c < 3; ++c)
+      dst[(long)c * HW] = bf2f(src[c]) / IMNET_STD[c];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Scaled squared-difference reduction (K18/K19):
+//   sum over logical elements of (255*(a-b))^2, fp64 accumulator.
+// flat variant (Clog == Cp) and channel-masked variant.
+// ---------------------------------------------------------------------------
+
+__global__ void k_sqdiff255_flat(const bf16_t* __restrict__ a,
+                                 const bf16_t* __restrict__ b,
+                                 double* __restrict__ out, long n) {
+  float s = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float d = 255.f * (bf2f(a[i]) - bf2f(b[i]));
+    s += d * d;
+  }
+  // wave reduce then one atomic per wave
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  if ((threadIdx.x & 63) == 0) atomicAdd(out, (double)s);
+}
+
+__global__ void k_sqdiff255_pix(const bf16_t* __restrict__ a,
+                                const bf16_t* __restrict__ b,
+                                double* __restrict__ out, long NHW, int Clog,
+                                int Cp) {
+  float s = 0.f;
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long o = p * Cp;
+    for (int c = 0; c < Clog; ++c) {
+      float d = 255.f * (bf2f(a[o + c]) - bf2f(b[o + c]));
+      s += d * d;
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  if ((threadIdx.x & 63) == 0) atomicAdd(out, (double)s);
+}
+
+// backward: da = gscale * (a - b), with gscale = grad * 2*255^2/numel
+__global__ void k_sqdiff255_bwd(const bf16_t* __restrict__ a,
+                                const bf16_t* __restrict__ b,
+                                const float* __restrict__ gscale,
+                                bf16_t* __restrict__ da, long n, float sign) {
+  const float gs = gscale[0] * sign;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    da[i] = f2bf(gs * (bf2f(a[i]) - bf2f(b[i])));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused flat Adam (K23): torch.optim.Adam semantics, fp32 master params.
+// ---------------------------------------------------------------------------
+
+// Graph-safe: the step counter and lr live in DEVICE buffers so a
+// hipGraph-captured step keeps correct bias correction and per-minibatch
+// StepLR semantics under replay (k_adam_tick runs first on the stream).
+__global__ void k_adam_tick(int* __restrict__ step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) step[0] += 1;
+}
+
+__global__ void k_adam(float* __restrict__ p, const float* __restrict__ g,
+                       float* __restrict__ m, float* __restrict__ v, long n,
+                       const float* __restrict__ lr_buf, float b1, float b2,
+                       float eps, const int* __restrict__ step) {
+  const float lr = lr_buf[0];
+  const float t = (float)step[0];
+  const float bc1 = 1.f - powf(b1, t);
+  const float bc2 = 1.f - powf(b2, t);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    float gi = g[i];
+    float mi = b1 * m[i] + (1.f - b1) * gi;
+    float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    float mhat = mi / bc1;
+    float vhat = vi / bc2;
+    p[i] -= lr * mhat / (sqrtf(vhat) + eps);
+  }
+}
+
+// ===========================================================================
+// Host wrappers
+// ===========================================================================
+
+std::vector<at::Tensor> build_inputs(const at::Tensor& raw,
+                                     const at::Tensor& wb,
+                                     const at::Tensor& ce,
+                                     const at::Tensor& gc) {
+  TORCH_CHECK(raw.is_cuda() && raw.dtype() == at::kFloat && raw.dim() == 4 &&
+              raw.size(1) == 3, "raw must be (N,3,H,W) fp32 CUDA");
+  const long N = raw.size(0), H = raw.size(2), W = raw.size(3);
+  const long NHW = N * H * W, HW = H * W;
+  auto opts = raw.options().dtype(at::kBFloat16);
+  auto cmg_in = at::empty({N, H, W, 16}, opts);
+  auto rwb = at::empty({N, H, W, 16}, opts);
+  auto rce = at::empty({N, H, W, 16}, opts);
+  auto rgc = at::empty({N, H, W, 16}, opts);
+  hipLaunchKernelGGL(k_build_inputs, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), raw.contiguous().data_ptr<float>(),
+                     wb.contiguous().data_ptr<float>(),
+                     ce.contiguous().data_ptr<float>(),
+                     gc.contiguous().data_ptr<float>(),
+                     (bf16_t*)cmg_in.data_ptr(), (bf16_t*)rwb.data_ptr(),
+                     (bf16_t*)rce.data_ptr(), (bf16_t*)rgc.data_ptr(), NHW,
+                     HW);
+  HIP_CHECK_LAST();
+  return {cmg_in, rwb, rce, rgc};
+}
+
+at::Tensor nchw_to_nhwc(const at::Tensor& x, int64_t Cp) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kFloat && x.dim() == 4);
+  const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  auto y = at::empty({N, H, W, Cp}, x.options().dtype(at::kBFloat16));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_nchw2nhwc, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), x.contiguous().data_ptr<float>(),
+                     (bf16_t*)y.data_ptr(), NHW, H * W, (int)C, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor nhwc_to_nchw(const at::Tensor& x, int64_t C) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 && x.dim() == 4);
+  const long N = x.size(0), H = x.size(1), W = x.size(2), Cp = x.size(3);
+  auto y = at::empty({N, C, H, W}, x.options().dtype(at::kFloat));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_nhwc2nchw, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)x.contiguous().data_ptr(),
+                     y.data_ptr<float>(), NHW, H * W, (int)C, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor fusion_fwd(const at::Tensor& maps, const at::Tensor& rwb,
+                      const at::Tensor& rce, const at::Tensor& rgc) {
+  const long N = maps.size(0), H = maps.size(1), W = maps.size(2);
+  const long NHW = N * H * W;
+  auto out = at::empty_like(maps);
+  hipLaunchKernelGGL(k_fusion_fwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)maps.data_ptr(),
+                     (const bf16_t*)rwb.data_ptr(),
+                     (const bf16_t*)rce.data_ptr(),
+                     (const bf16_t*)rgc.data_ptr(), (bf16_t*)out.data_ptr(),
+                     NHW);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+std::vector<at::Tensor> fusion_bwd(const at::Tensor& dout,
+                                   const at::Tensor& maps,
+                                   const at::Tensor& rwb,
+                                   const at::Tensor& rce,
+                                   const at::Tensor& rgc) {
+  const long N = maps.size(0), H = maps.size(1), W = maps.size(2);
+  const long NHW = N * H * W;
+  auto dmaps = at::empty_like(maps);
+  auto drwb = at::empty_like(maps);
+  auto drce = at::empty_like(maps);
+  auto drgc = at::empty_like(maps);
+  hipLaunchKernelGGL(k_fusion_bwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)dout.data_ptr(),
+                     (const bf16_t*)maps.data_ptr(),
+                     (const bf16_t*)rwb.data_ptr(),
+                     (const bf16_t*)rce.data_ptr(),
+                     (const bf16_t*)rgc.data_ptr(), (bf16_t*)dmaps.data_ptr(),
+                     (bf16_t*)drwb.data_ptr(), (bf16_t*)drce.data_ptr(),
+                     (bf16_t*)drgc.data_ptr(), NHW);
+  HIP_CHECK_LAST();
+  return {dmaps, drwb, drce, drgc};
+}
+
+at::Tensor act_bwd(const at::Tensor& dy, const at::Tensor& y, int64_t act) {
+  auto dpre = at::empty_like(dy);
+  const long n = dy.numel();
+  hipLaunchKernelGGL(k_act_bwd, dim3(grid1d(n)), dim3(TPB), 0, cur_stream(),
+                     (const bf16_t*)dy.data_ptr(),
+                     (const bf16_t*)y.data_ptr(), (bf16_t*)dpre.data_ptr(),
+                     n, (int)act);
+  HIP_CHECK_LAST();
+  return dpre;
+}
+
+at::Tensor normalize_vgg_fwd(const at::Tensor& x, int64_t Cp) {
+  const long N = x.size(0), H = x.size(2), W = x.size(3);
+  const long NHW = N * H * W;
+  auto y = at::empty({N, H, W, Cp}, x.options().dtype(at::kBFloat16));
+  hipLaunchKernelGGL(k_normalize_vgg_fwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), x.contiguous().data_ptr<float>(),
+                     (bf16_t*)y.data_ptr(), NHW, H * W, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor normalize_vgg_bwd(const at::Tensor& dy, int64_t C) {
+  TORCH_CHECK(C == 3);
+  const long N = dy.size(0), H = dy.size(1), W = dy.size(2),
+             Cp = dy.size(3);
+  auto dx = at::empty({N, C, H, W}, dy.options().dtype(at::kFloat));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_normalize_vgg_bwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)dy.contiguous().data_ptr(),
+                     dx.data_ptr<float>(), NHW, H * W, (int)Cp);
+  HIP_CHECK_LAST();
+  return dx;
+}
+
+at::Tensor sqdiff255_sum(const at::Tensor& a, const at::Tensor& b,
+                         int64_t Clog) {
+  TORCH_CHECK(a.sizes() == b.sizes());
+  auto out = at::zeros({}, a.options().dtype(at::kDouble));
+  const long Cp = a.size(3);
+  if (Clog == Cp) {
+    const long n = a.numel();
+    hipLaunchKernelGGL(k_sqdiff255_flat, dim3(grid1d(n)), dim3(TPB), 0,
+                       cur_stream(), (const bf16_t*)a.data_ptr(),
+                       (const bf16_t*)b.data_ptr(), out.data_ptr<double>(),
+                       n);
+  } else {
+    const long NHW = a.numel() / Cp;
+    hipLaunchKernelGGL(k_sqdiff255_pix, dim3(grid1d(NHW)), dim3(TPB), 0,
+                       cur_stream(), (const bf16_t*)a.data_ptr(),
+                       (const bf16_t*)b.data_ptr(), out.data_ptr<double>(),
+                       NHW, (int)Clog, (int)Cp);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}
+
+at::Tensor sqdiff255_bwd(const at::Tensor& a, const at::Tensor& b,
+                         const at::Tensor& gscale, double sign) {
+  auto da = at::empty_like(a);
+  const long n = a.numel();
+  hipLaunchKernelGGL(k_sqdiff255_bwd, dim3(grid1d(n)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)a.data_ptr(),
+                     (const bf16_t*)b.data_ptr(), gscale.data_ptr<float>(),
+                     (bf16_t*)da.data_ptr(), n, (float)sign);
+  HIP_CHECK_LAST();
+  return da;
+}
+
+void adam_step(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
+               at::Tensor& v, const at::Tensor& lr_buf, double b1, double b2,
+               double eps, at::Tensor& step_buf) {
+  TORCH_CHECK(p.is_cuda() && p.dtype() == at::kFloat);
+  TORCH_CHECK(lr_buf.dtype() == at::kFloat && step_buf.dtype() == at::kInt);
+  const long n = p.numel();
+  hipLaunchKernelGGL(k_adam_tick, dim3(1), dim3(64), 0, cur_stream(),
+                     step_buf.data_ptr<int>());
+  hipLaunchKernelGGL(k_adam, dim3(grid1d(n)), dim3(TPB), 0, cur_stream(),
+                     p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), n,
+                     lr_buf.data_ptr<float>(), (float)b1, (float)b2,
+                     (float)eps, step_buf.data_ptr<int>());
+  HIP_CHECK_LAST();
+}

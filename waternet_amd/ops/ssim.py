@@ -1,0 +1,20 @@
+"""Native HIP SSIM (metric path, no autograd)."""
+
+import torch
+
+from waternet_amd.ops import ext
+
+
+def ssim_native(preds, target, data_range, kernel_size=11, sigma=1.5,
+                k1=0.01, k2=0.03):
+    if kernel_size != 11 or abs(sigma - 1.5) > 1e-9:
+        from waternet_amd.utils.metrics import _ssim_torch
+
+        return _ssim_torch(preds, target, data_range, kernel_size, sigma, k1,
+                           k2)
+    a = preds.float().contiguous()
+    b = target.float().contiguous()
+    s = ext().ssim_sum(a, b, float(data_range), k1, k2)
+    n, c, h, w = a.shape
+    count = n * c * (h - kernel_size + 1) * (w - kernel_size + 1)
+    return (s / count).to(torch.float32)
