@@ -68,9 +68,32 @@ def load_model(weights, device):
     return model
 
 
+_ENGINES = {}
+
+
+def _gpu_engine(model, h, w, device):
+    key = (id(model), h, w)
+    eng = _ENGINES.get(key)
+    if eng is None:
+        from waternet_amd.engine.inferencer import InferenceEngine
+
+        eng = InferenceEngine(model, h, w, device=device)
+        _ENGINES[key] = eng
+    return eng
+
+
 @torch.no_grad()
 def enhance_frame(model, rgb: np.ndarray, device) -> np.ndarray:
-    """uint8 HWC RGB -> enhanced uint8 HWC RGB."""
+    """uint8 HWC RGB -> enhanced uint8 HWC RGB. On GPU the whole frame
+    pipeline (preprocess + forward + postprocess) runs as a hipGraph-
+    captured on-device pipeline; on CPU the reference-semantics numpy
+    transforms + eager model run."""
+    if device.type == "cuda" and rgb.shape[0] % 8 == 0 and rgb.shape[1] % 8 == 0:
+        from waternet_amd.ops import native_available
+
+        if native_available():
+            eng = _gpu_engine(model, rgb.shape[0], rgb.shape[1], device)
+            return eng.infer_frame(rgb)
     wb, gc, he = transform(rgb)
     rgb_ten = arr2ten(rgb, add_batch_dim=True).to(device)
     wb_ten = arr2ten(wb, add_batch_dim=True).to(device)

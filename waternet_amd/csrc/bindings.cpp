@@ -34,6 +34,8 @@ at::Tensor sqdiff255_sum(const at::Tensor& a, const at::Tensor& b,
                          int64_t Clog);
 at::Tensor sqdiff255_bwd(const at::Tensor& a, const at::Tensor& b,
                          const at::Tensor& gscale, double sign);
+at::Tensor out_to_u8(const at::Tensor& x);
+at::Tensor u8_to_nchw(const at::Tensor& x);
 void adam_step(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
                at::Tensor& v, const at::Tensor& lr_buf, double b1, double b2,
                double eps, at::Tensor& step_buf);
@@ -68,6 +70,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sqdiff255_sum", &sqdiff255_sum);
   m.def("sqdiff255_bwd", &sqdiff255_bwd);
   m.def("adam_step", &adam_step);
+  m.def("out_to_u8", &out_to_u8);
+  m.def("u8_to_nchw", &u8_to_nchw);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
   m.def("ssim_sum", &ssim_sum);

@@ -287,6 +287,37 @@ __global__ void k_adam(float* __restrict__ p, const float* __restrict__ g,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused postprocess (K27): NHWC bf16 model output -> uint8 RGB
+// (clip to [0,1], *255, truncate — ten2arr semantics training_utils.py:27-43)
+// and uint8 NHWC -> NCHW fp32 [0,1] input bridge.
+// ---------------------------------------------------------------------------
+
+__global__ void k_out_to_u8(const bf16_t* __restrict__ x,
+                            uint8_t* __restrict__ y, long NHW, int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+#pragma unroll
+    for (int c = 0; c < 3; ++c) {
+      float v = bf2f(x[p * Cp + c]);
+      v = fminf(fmaxf(v, 0.f), 1.f) * 255.f;
+      y[p * 3 + c] = (uint8_t)v;
+    }
+  }
+}
+
+__global__ void k_u8_to_nchw(const uint8_t* __restrict__ x,
+                             float* __restrict__ y, long NHW, long HW) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long n = p / HW, rem = p - n * HW;
+    float* dst = y + (n * 3) * HW + rem;
+#pragma unroll
+    for (int c = 0; c < 3; ++c)
+      dst[(long)c * HW] = (float)x[p * 3 + c] * (1.f / 255.f);
+  }
+}
+
 // ===========================================================================
 // Host wrappers
 // ===========================================================================
@@ -445,6 +476,31 @@ at::Tensor sqdiff255_bwd(const at::Tensor& a, const at::Tensor& b,
                      (bf16_t*)da.data_ptr(), n, (float)sign);
   HIP_CHECK_LAST();
   return da;
+}
+
+at::Tensor out_to_u8(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 && x.dim() == 4);
+  const long N = x.size(0), H = x.size(1), W = x.size(2), Cp = x.size(3);
+  auto y = at::empty({N, H, W, 3}, x.options().dtype(at::kByte));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_out_to_u8, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)x.contiguous().data_ptr(),
+                     y.data_ptr<uint8_t>(), NHW, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor u8_to_nchw(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kByte && x.dim() == 4 &&
+              x.size(3) == 3);
+  const long N = x.size(0), H = x.size(1), W = x.size(2);
+  auto y = at::empty({N, 3, H, W}, x.options().dtype(at::kFloat));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_u8_to_nchw, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), x.contiguous().data_ptr<uint8_t>(),
+                     y.data_ptr<float>(), NHW, H * W);
+  HIP_CHECK_LAST();
+  return y;
 }
 
 void adam_step(at::Tensor& p, const at::Tensor& g, at::Tensor& m,

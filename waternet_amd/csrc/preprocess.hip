@@ -78,21 +78,24 @@ __global__ void k_wb_params(const unsigned int* __restrict__ hist,
                             float* __restrict__ params,  // (N,3,2) lo,hi
                             long HW) {
   const long n = blockIdx.x;
-  const int ch = threadIdx.x;  // 3 threads do work
-  if (ch >= 3) return;
-  const unsigned int* h = hist + (n * 3 + ch) * 256;
+  const int ch = threadIdx.x;  // 3 lanes do work; no early return (barrier)
   __shared__ double sums[3];
-  unsigned long long s = 0;
-  for (int v = 0; v < 256; ++v) s += (unsigned long long)h[v] * v;
-  sums[ch] = (double)s;
+  const unsigned int* h = hist + (n * 3 + min(ch, 2)) * 256;
+  if (ch < 3) {
+    unsigned long long s = 0;
+    for (int v = 0; v < 256; ++v) s += (unsigned long long)h[v] * v;
+    sums[ch] = (double)s;
+  }
   __syncthreads();
-  double maxsum = fmax(sums[0], fmax(sums[1], sums[2]));
-  double ratio = maxsum / sums[ch];
-  double sat = 0.005 * ratio;
-  double lo = quantile_from_hist(h, HW, sat);
-  double hi = quantile_from_hist(h, HW, 1.0 - sat);
-  params[(n * 3 + ch) * 2 + 0] = (float)lo;
-  params[(n * 3 + ch) * 2 + 1] = (float)hi;
+  if (ch < 3) {
+    double maxsum = fmax(sums[0], fmax(sums[1], sums[2]));
+    double ratio = maxsum / sums[ch];
+    double sat = 0.005 * ratio;
+    double lo = quantile_from_hist(h, HW, sat);
+    double hi = quantile_from_hist(h, HW, 1.0 - sat);
+    params[(n * 3 + ch) * 2 + 0] = (float)lo;
+    params[(n * 3 + ch) * 2 + 1] = (float)hi;
+  }
 }
 
 // ---------------------------------------------------------------------------
