@@ -66,37 +66,45 @@ class BenchTrainer:
 
     # ---- one full training step on the current stream ----
     def _run_step_body(self):
+        from waternet_amd.utils.profiling import trace_range
+
         raw_u8, ref_u8 = self.raw_static, self.ref_static
-        wb_u8, gc_u8, he_u8 = gpu_transform_batch(raw_u8)
+        with trace_range("preprocess"):
+            wb_u8, gc_u8, he_u8 = gpu_transform_batch(raw_u8)
 
-        def to_f(t_u8):
-            return t_u8.permute(0, 3, 1, 2).float().div_(255.0)
+            from waternet_amd.ops import ext
 
-        raw_f = to_f(raw_u8)
-        wb_f = to_f(wb_u8)
-        gc_f = to_f(gc_u8)
-        he_f = to_f(he_u8)
-        ref_f = to_f(ref_u8)
+            e = ext()
+            raw_f = e.u8_to_nchw(raw_u8)
+            wb_f = e.u8_to_nchw(wb_u8)
+            gc_f = e.u8_to_nchw(gc_u8)
+            he_f = e.u8_to_nchw(he_u8)
+            ref_f = e.u8_to_nchw(ref_u8)
 
-        out = self.model(raw_f, wb_f, he_f, gc_f)  # he in the ce slot
+        with trace_range("forward"):
+            out = self.model(raw_f, wb_f, he_f, gc_f)  # he in the ce slot
 
-        fx = self.vgg(normalize_imagenet(out))
-        with torch.no_grad():
-            fy = self.vgg(normalize_imagenet(ref_f))
-        dp = 255.0 * (fx - fy)
-        perceptual = torch.mean(dp * dp)
-        dm = 255.0 * (out - ref_f)
-        mse = torch.mean(dm * dm)
-        loss = PERCEPTUAL_WEIGHT * perceptual + mse
+        with trace_range("loss"):
+            fx = self.vgg(normalize_imagenet(out))
+            with torch.no_grad():
+                fy = self.vgg(normalize_imagenet(ref_f))
+            dp = 255.0 * (fx - fy)
+            perceptual = torch.mean(dp * dp)
+            dm = 255.0 * (out - ref_f)
+            mse = torch.mean(dm * dm)
+            loss = PERCEPTUAL_WEIGHT * perceptual + mse
 
-        self.opt.zero_grad()
-        loss.backward()
+        with trace_range("backward"):
+            self.opt.zero_grad()
+            loss.backward()
         if self.world > 1:
-            self.opt.grads.div_(self.world)
-            torch.distributed.all_reduce(self.opt.grads)
-        self.opt.step()
+            with trace_range("allreduce"):
+                self.opt.grads.div_(self.world)
+                torch.distributed.all_reduce(self.opt.grads)
+        with trace_range("optimizer"):
+            self.opt.step()
 
-        with torch.no_grad():
+        with trace_range("metrics"), torch.no_grad():
             ssim = ssim_native(out.detach(), ref_f, 1.0)
             mse01 = torch.mean((out.detach() - ref_f) ** 2)
             psnr = 10.0 * torch.log10(1.0 / mse01)
