@@ -49,13 +49,14 @@ WN_DEVFN constexpr int frag_k16(int g, int e) {
 // Forward / dgrad implicit GEMM
 // ---------------------------------------------------------------------------
 
-template <int BN, int KS>
+template <int BN, int KS, bool SPLITK = false>
 __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     const bf16_t* __restrict__ X,   // (N, H, W, Cp)
     const bf16_t* __restrict__ Wp,  // [Kp][RS*Cp] k-major packed
     const float* __restrict__ Bias, // [>=Klog] or nullptr
     bf16_t* __restrict__ Y,         // (N, H, W, Kp)
-    int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act) {
+    int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act,
+    float* __restrict__ Y32 = nullptr) {  // SPLITK: fp32 partials (pre-zeroed)
   constexpr int BM = 128;
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;
@@ -204,17 +205,25 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
 #pragma unroll
     for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int nk = (KG + 31) / 32;
-  loadA(0);
-  loadB(0);
+  const int nkAll = (KG + 31) / 32;
+  // SPLITK: z-slice [ks0, nk) of the K loop; else the whole range.
+  int ks0 = 0, nk = nkAll;
+  if (SPLITK) {
+    const int per = (nkAll + gridDim.z - 1) / gridDim.z;
+    ks0 = blockIdx.z * per;
+    nk = min(nkAll, ks0 + per);
+    if (ks0 >= nk) return;
+  }
+  loadA(ks0 * 32);
+  loadB(ks0 * 32);
   writeTiles(0);
   __syncthreads();
 
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
-  for (int ks = 0; ks < nk; ++ks) {
-    const int cur = ks & 1;
+  for (int ks = ks0; ks < nk; ++ks) {
+    const int cur = (ks - ks0) & 1;
     if (ks + 1 < nk) {
       loadA((ks + 1) * 32);
       loadB((ks + 1) * 32);
@@ -245,7 +254,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     }
   }
 
-  // ---- epilogue: bias + activation (+ pad-channel zeroing) ----
+  // ---- epilogue: bias + activation (+ pad-channel zeroing), or fp32
+  //      partial-sum atomics when this launch is a split-K slice ----
   const int lr4 = (lane >> 4) * 4;
 #pragma unroll
   for (int fn = 0; fn < FN; ++fn) {
@@ -259,149 +269,229 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
       for (int r = 0; r < 4; ++r) {
         long m = m0 + (wr * FM + fm) * 16 + lr4 + r;
         if (m >= M) continue;
-        float v = acc[fm][fn][r] + bv;
-        if (act == ACT_RELU) v = fmaxf(v, 0.f);
-        else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
-        if (kpad) v = 0.f;
-        Y[m * Kp + k] = f2bf(v);
+        if (SPLITK) {
+          atomicAdd(&Y32[m * Kp + k], acc[fm][fn][r]);
+        } else {
+          float v = acc[fm][fn][r] + bv;
+          if (act == ACT_RELU) v = fmaxf(v, 0.f);
+          else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
+          if (kpad) v = 0.f;
+          Y[m * Kp + k] = f2bf(v);
+        }
       }
     }
   }
 }
 
+// Split-K finalize: Y = act(Y32 + bias), pad channels zeroed.
+__global__ void k_splitk_finalize(const float* __restrict__ Y32,
+                                  const float* __restrict__ Bias,
+                                  bf16_t* __restrict__ Y, long M, int Kp,
+                                  int Klog, int act) {
+  const long total = M * Kp;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(i % Kp);
+    float v = Y32[i] + ((Bias != nullptr && k < Klog) ? Bias[k] : 0.f);
+    if (act == ACT_RELU) v = fmaxf(v, 0.f);
+    else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
+    if (k >= Klog) v = 0.f;
+    Y[i] = f2bf(v);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Weight gradient: dW[k][rsc] = sum_m dY[m][k] * X[m][rsc]
-// 64x64 output tile per block, K-dim = m in steps of 32, fp32 atomics into
-// the NCHW fp32 grad tensor.
+//
+// v2 (tuned from rocprofv3 r01 stats where wgrad was 33% of step time):
+//   - BK x 128 output tile per block (BK = min(Kp,128)), 2x2 / 1x4 waves
+//   - m-chunks of 64 (two MFMA k-steps); single LDS buffer with hoisted
+//     fragment reads: per chunk, all ds_read_b128 fragments for the chunk
+//     are pulled into registers first, then after one barrier the staging
+//     writes for the NEXT chunk issue back-to-back with the chunk's MFMAs
+//     (independent, so the LDS writes hide under matrix-core work).
+//   - LDS is row-major [row][m] with +8 element row padding: staging writes
+//     come from contiguous-k global vector loads (bf16x8) and scatter 8
+//     ds_write_b16; fragment reads are aligned ds_read_b128 with 36-dword
+//     row stride (16 distinct banks, conflict-free).
+//   - split-m z dimension sized to fill 256 CUs x 2 blocks; gy (rsc tiles)
+//     varies fastest so co-resident blocks share dY/X chunks through L2.
+//   - fp32 atomicAdd epilogue into the NCHW fp32 grad tensor (pre-zeroed).
+// Requires the KMAP-0 fragment layout (frag_k16(g,e) = g*8+e).
 // ---------------------------------------------------------------------------
 
-template <int KS>
+template <int KS, int BK>
 __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     const bf16_t* __restrict__ dY,  // (N,H,W,Kp)
     const bf16_t* __restrict__ X,   // (N,H,W,Cp)
     float* __restrict__ dW,         // (K, C, KS, KS) fp32, pre-zeroed
     int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C,
     int splitm) {
+  static_assert(WN_MFMA_KMAP == 0, "wgrad staging assumes KMAP 0");
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;
+  constexpr int BR = 128;           // rsc tile
+  constexpr int CH = 64;            // m chunk
+  constexpr int LROW = CH + 8;      // padded LDS row (elements)
+  constexpr int WR = (BK >= 64) ? 2 : 1;   // wave rows (k dim)
+  constexpr int WC = 4 / WR;               // wave cols (rsc dim)
+  constexpr int FK = BK / WR / 16;  // k fragments per wave
+  constexpr int FR = BR / WC / 16;  // rsc fragments per wave
+  constexpr int ATOT = CH * BK / 8;         // total A vec8 slots (>=128)
+  constexpr int ASLOT = (ATOT + 255) / 256; // A vec8 loads per thread
+  constexpr int BSLOT = CH * BR / 8 / 256;  // B vec8 loads per thread
   const int KG = RS * Cp;
   const long M = (long)N * H * W;
   const int HW = H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);  // 64k x 32m frag-major
-  bf16_t* lB = lA + 64 * 32;                     // 64rsc x 32m frag-major
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // BK x LROW
+  bf16_t* lB = lA + BK * LROW;                    // BR x LROW
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
-  const int wr = wid >> 1;  // 2x2 waves, 32x32 tile each
-  const int wc = wid & 1;
-  const int kt0 = blockIdx.x * 64;
-  const int rt0 = blockIdx.y * 64;
+  const int wr = wid / WC;
+  const int wc = wid % WC;
+  const int kt0 = blockIdx.x * BK;
+  const int rt0 = blockIdx.y * BR;
 
-  // staging geometry: thread -> (mrow = tid>>3 in 0..31, grp = tid&7)
-  const int mrow = tid >> 3;
-  const int grp = tid & 7;
-
-  f32x4 acc[2][2];
+  // ---- static staging geometry ----
+  // A slot s: jg = slot % (BK/8) (k group), ml = slot / (BK/8) (m row)
+  int aJg[ASLOT], aMl[ASLOT];
+  bool aV[ASLOT];
 #pragma unroll
-  for (int a = 0; a < 2; ++a)
+  for (int s = 0; s < ASLOT; ++s) {
+    int slot = tid + s * 256;
+    aV[s] = slot < ATOT;
+    slot = aV[s] ? slot : 0;
+    aJg[s] = slot % (BK / 8);
+    aMl[s] = slot / (BK / 8);
+  }
+  // B slot s: rsc group + m row; decode tap/c once (fixed per thread)
+  int bMl[BSLOT], bC[BSLOT], bDy[BSLOT], bDx[BSLOT];
+  bool bIn[BSLOT];
 #pragma unroll
-    for (int b = 0; b < 2; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int s = 0; s < BSLOT; ++s) {
+    int slot = tid + s * 256;
+    int jg = slot % (BR / 8);
+    bMl[s] = slot / (BR / 8);
+    int rsc = rt0 + jg * 8;
+    bIn[s] = rsc < KG;
+    int tap = rsc >> log2Cp;
+    bC[s] = rsc & (Cp - 1);
+    bDy[s] = tap / KS;
+    bDx[s] = tap - bDy[s] * KS;
+  }
 
-  const long nChunks = (M + 31) / 32;
-  for (long chunk = blockIdx.z; chunk < nChunks; chunk += splitm) {
-    const long m0 = chunk * 32;
-    const long m = m0 + mrow;
-    const bool mv = m < M;
+  bf16x8 aR[ASLOT], bR[BSLOT];
 
-    // ---- stage A (dY^T): load 8 k for (m, kgrp), scatter-write ----
-    {
+  auto loadA = [&](long m0) {
+#pragma unroll
+    for (int s = 0; s < ASLOT; ++s) {
       bf16x8 v = {};
-      int kg = kt0 + grp * 8;
-      if (mv && kg + 7 < Kp)
-        v = *reinterpret_cast<const bf16x8*>(dY + m * Kp + kg);
-      else if (mv) {
-        for (int e = 0; e < 8; ++e)
-          if (kg + e < Kp) v[e] = dY[m * Kp + kg + e];
+      long m = m0 + aMl[s];
+      if (m < M) {
+        int kg = kt0 + aJg[s] * 8;
+        if (kg + 7 < Kp)
+          v = *reinterpret_cast<const bf16x8*>(dY + m * Kp + kg);
       }
-      const int kb = mrow >> 3, eM = mrow & 7;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int kl = grp * 8 + e;  // local k 0..63
-        int i = kl & 15, kf = kl >> 4;
-        // element eM must sit at position where frag_k16(kb, pos) == mrow&..:
-        // frag element index epos satisfies frag_k16(kb, epos) = mrow for the
-        // reading lane group kb' = reading lane>>4. Store m-major into the
-        // frag slot: slot (kf, kb, i), element epos with
-        // frag_k16(kb, epos) == mrow (within 0..31).
-        lA[((kf * 4 + kb) * 16 + i) * 8 + eM] = v[e];
-      }
+      aR[s] = v;
     }
-    // ---- stage B (X gather, transposed the same way) ----
-    {
+  };
+  auto loadB = [&](long m0) {
+#pragma unroll
+    for (int s = 0; s < BSLOT; ++s) {
       bf16x8 v = {};
-      if (mv) {
-        long mm = m;
-        int n = (int)(mm / HW);
-        int rem = (int)(mm - (long)n * HW);
+      long m = m0 + bMl[s];
+      if (m < M && bIn[s]) {
+        int n = (int)(m / HW);
+        int rem = (int)(m - (long)n * HW);
         int oy = rem / W, ox = rem - (rem / W) * W;
-        int rsc = rt0 + grp * 8;
-        if (rsc < KG) {
-          int tap = rsc >> log2Cp;
-          int c = rsc & (Cp - 1);
-          int dy = tap / KS, dx = tap - (tap / KS) * KS;
-          int iy = oy + dy - PAD, ix = ox + dx - PAD;
-          if (iy >= 0 && iy < H && ix >= 0 && ix < W)
-            v = *reinterpret_cast<const bf16x8*>(
-                X + (((long)(n * H + iy) * W + ix) << log2Cp) + c);
-        }
+        int iy = oy + bDy[s] - PAD, ix = ox + bDx[s] - PAD;
+        if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+          v = *reinterpret_cast<const bf16x8*>(
+              X + (((long)(n * H + iy) * W + ix) << log2Cp) + bC[s]);
       }
-      const int kb = mrow >> 3, eM = mrow & 7;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int jl = grp * 8 + e;
-        int j = jl & 15, jf = jl >> 4;
-        lB[((jf * 4 + kb) * 16 + j) * 8 + eM] = v[e];
-      }
+      bR[s] = v;
     }
-    __syncthreads();
+  };
+  auto writeTiles = [&]() {
+#pragma unroll
+    for (int s = 0; s < ASLOT; ++s) {
+      if (!aV[s]) continue;
+      bf16_t* dst = lA + aJg[s] * 8 * LROW + aMl[s];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e * LROW] = aR[s][e];
+    }
+#pragma unroll
+    for (int s = 0; s < BSLOT; ++s) {
+      int slot = tid + s * 256;
+      int jg = slot % (BR / 8);
+      bf16_t* dst = lB + jg * 8 * LROW + bMl[s];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e * LROW] = bR[s][e];
+    }
+  };
 
-    // ---- MFMA: K-dim = 32 m values ----
-    const int lg = lane >> 4, li = lane & 15;
-    bf16x8 aF[2], bF[2];
+  f32x4 acc[FK][FR];
 #pragma unroll
-    for (int f = 0; f < 2; ++f) {
-      // NOTE: LDS holds elements m-ordered (eM = m&7 at position eM), but the
-      // MFMA consumes element e as k = frag_k16(g, e). For KMAP 0 these agree
-      // (frag_k16(g,e) = g*8+e and we stored m = kb*8+eM at slot(kb)+eM).
-      // For KMAP 1 the staging writes must permute: handled below.
-      aF[f] = *reinterpret_cast<const bf16x8*>(
-          lA + (((wr * 2 + f) * 4 + lg) * 16 + li) * 8);
-      bF[f] = *reinterpret_cast<const bf16x8*>(
-          lB + (((wc * 2 + f) * 4 + lg) * 16 + li) * 8);
-    }
+  for (int a = 0; a < FK; ++a)
 #pragma unroll
-    for (int fa = 0; fa < 2; ++fa)
-#pragma unroll
-      for (int fb = 0; fb < 2; ++fb)
-        acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            aF[fa], bF[fb], acc[fa][fb], 0, 0, 0);
+    for (int b = 0; b < FR; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const long nChunks = (M + CH - 1) / CH;
+  const int lg = lane >> 4, li = lane & 15;
+
+  long chunk = blockIdx.z;
+  if (chunk < nChunks) {
+    loadA(chunk * CH);
+    loadB(chunk * CH);
+    writeTiles();
     __syncthreads();
+  }
+  for (; chunk < nChunks; chunk += splitm) {
+    const long next = chunk + splitm;
+    if (next < nChunks) {
+      loadA(next * CH);  // global -> regs, overlaps this chunk's math
+      loadB(next * CH);
+    }
+    // hoist ALL fragment reads for this chunk into registers
+    bf16x8 aF[CH / 32][FK], bF[CH / 32][FR];
+#pragma unroll
+    for (int ks32 = 0; ks32 < CH / 32; ++ks32) {
+#pragma unroll
+      for (int f = 0; f < FK; ++f)
+        aF[ks32][f] = *reinterpret_cast<const bf16x8*>(
+            lA + ((wr * FK + f) * 16 + li) * LROW + ks32 * 32 + lg * 8);
+#pragma unroll
+      for (int f = 0; f < FR; ++f)
+        bF[ks32][f] = *reinterpret_cast<const bf16x8*>(
+            lB + ((wc * FR + f) * 16 + li) * LROW + ks32 * 32 + lg * 8);
+    }
+    __syncthreads();  // all waves done reading; LDS free for restaging
+    if (next < nChunks) writeTiles();  // issues alongside the MFMAs below
+#pragma unroll
+    for (int ks32 = 0; ks32 < CH / 32; ++ks32)
+#pragma unroll
+      for (int fa = 0; fa < FK; ++fa)
+#pragma unroll
+        for (int fb = 0; fb < FR; ++fb)
+          acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aF[ks32][fa], bF[ks32][fb], acc[fa][fb], 0, 0, 0);
+    if (next < nChunks) __syncthreads();  // writes visible before next reads
   }
 
   // ---- epilogue: scatter-add into NCHW fp32 dW ----
-  const int lg = lane >> 4, li = lane & 15;
 #pragma unroll
-  for (int fa = 0; fa < 2; ++fa) {
+  for (int fa = 0; fa < FK; ++fa) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int k = kt0 + (wr * 2 + fa) * 16 + lg * 4 + r;
+      int k = kt0 + (wr * FK + fa) * 16 + lg * 4 + r;
       if (k >= K) continue;
 #pragma unroll
-      for (int fb = 0; fb < 2; ++fb) {
-        int rsc = rt0 + (wc * 2 + fb) * 16 + li;
+      for (int fb = 0; fb < FR; ++fb) {
+        int rsc = rt0 + (wc * FR + fb) * 16 + li;
         if (rsc >= KG) continue;
         int tap = rsc >> log2Cp;
         int c = rsc & (Cp - 1);
@@ -507,16 +597,38 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
   const float* bptr =
       bias.has_value() ? bias->data_ptr<float>() : nullptr;
   const int gx = (int)((M + 127) / 128);
+  const int nk = (KS * KS * Cp + 31) / 32;
 
   auto launch = [&](auto bn_const) {
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
     const size_t lds = (2 * 128 * 32 + 2 * BN * 32) * sizeof(bf16_t);
-    hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256), lds,
-                       stream, (const bf16_t*)x.data_ptr(),
-                       (const bf16_t*)wp.data_ptr(), bptr,
-                       (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
-                       Klog, act);
+    // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
+    // 256 CUs idle; split the K loop across gz slices into fp32 partials,
+    // then finalize bias+act+bf16 in a second tiny pass.
+    const int base = gx * gy;
+    int gz = 1;
+    if (base < 320 && nk >= 16) gz = std::min(std::max(1, 512 / base), nk / 4);
+    if (gz > 1) {
+      auto y32 = at::zeros({M, (long)Kp}, x.options().dtype(at::kFloat));
+      hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
+                         dim3(256), lds, stream,
+                         (const bf16_t*)x.data_ptr(),
+                         (const bf16_t*)wp.data_ptr(), bptr,
+                         (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
+                         Klog, act, y32.data_ptr<float>());
+      const long total = M * Kp;
+      const int fb = (int)std::min<long>(1024, (total + 255) / 256);
+      hipLaunchKernelGGL(k_splitk_finalize, dim3(fb), dim3(256), 0, stream,
+                         y32.data_ptr<float>(), bptr,
+                         (bf16_t*)y.data_ptr(), M, Kp, Klog, act);
+    } else {
+      hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256),
+                         lds, stream, (const bf16_t*)x.data_ptr(),
+                         (const bf16_t*)wp.data_ptr(), bptr,
+                         (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
+                         Klog, act);
+    }
   };
   if (Kp >= 128)
     launch(std::integral_constant<int, 128>{});
@@ -571,25 +683,37 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   const int K = dw.size(0), C = dw.size(1);
   TORCH_CHECK(dw.size(2) == ks && dw.size(3) == ks);
   const int KG = (int)(ks * ks) * Cp;
-  const int gx = (Kp + 63) / 64, gy = (KG + 63) / 64;
-  // pick split so gx*gy*split ~ 2*256 blocks
-  int split = std::max(1, 512 / std::max(1, gx * gy));
-  const long nChunks = ((long)N * H * W + 31) / 32;
+  const int BK = std::min(Kp, 128);
+  const int gx = (Kp + BK - 1) / BK, gy = (KG + 127) / 128;
+  // split so gx*gy*split fills 256 CUs x 2 resident blocks
+  int split = std::max(1, 640 / std::max(1, gx * gy));
+  const long nChunks = ((long)N * H * W + 63) / 64;
   split = (int)std::min<long>(split, nChunks);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
-  const size_t lds = 2 * 64 * 32 * sizeof(bf16_t);
-  auto launch = [&](auto ks_const) {
+  const size_t lds = (size_t)(BK + 128) * 72 * sizeof(bf16_t);
+  auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
-    hipLaunchKernelGGL((k_conv_wgrad<KSV>), dim3(gx, gy, split), dim3(256),
-                       lds, stream, (const bf16_t*)dy.data_ptr(),
+    constexpr int BKV = decltype(bk_const)::value;
+    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV>), dim3(gx, gy, split),
+                       dim3(256), lds, stream, (const bf16_t*)dy.data_ptr(),
                        (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
                        H, W, Cp, log2i(Cp), Kp, K, C, split);
   };
+  auto launch_ks = [&](auto ks_const) {
+    if (BK == 128)
+      launch(ks_const, std::integral_constant<int, 128>{});
+    else if (BK == 64)
+      launch(ks_const, std::integral_constant<int, 64>{});
+    else if (BK == 32)
+      launch(ks_const, std::integral_constant<int, 32>{});
+    else
+      launch(ks_const, std::integral_constant<int, 16>{});
+  };
   switch (ks) {
-    case 1: launch(std::integral_constant<int, 1>{}); break;
-    case 3: launch(std::integral_constant<int, 3>{}); break;
-    case 5: launch(std::integral_constant<int, 5>{}); break;
-    case 7: launch(std::integral_constant<int, 7>{}); break;
+    case 1: launch_ks(std::integral_constant<int, 1>{}); break;
+    case 3: launch_ks(std::integral_constant<int, 3>{}); break;
+    case 5: launch_ks(std::integral_constant<int, 5>{}); break;
+    case 7: launch_ks(std::integral_constant<int, 7>{}); break;
     default: TORCH_CHECK(false, "unsupported kernel size ", ks);
   }
   HIP_CHECK_LAST();
