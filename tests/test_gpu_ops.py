@@ -77,6 +77,23 @@ def test_conv_fwd_sigmoid(ext):
     assert y[..., K:].abs().max().item() == 0.0  # sigmoid pad masked to 0
 
 
+def test_tr_read_lane_mapping(ext):
+    """ds_read_b64_tr_b16 must deliver column (lane&15), rows mblk*4..+3 of
+    the [4][16] subtile addressed by the wgrad tr image layout — the wgrad
+    kernel's fragment correctness rests on exactly this mapping."""
+    out = ext.probe_tr().cpu()  # (64, 2, 8): lane, col-block, elem
+    for lane in range(64):
+        g, i = lane >> 4, lane & 15
+        for f in range(2):
+            for j in range(8):
+                m = 8 * g + j if j < 4 else 8 * g + (j - 4) + 4
+                col = f * 16 + i
+                expect = float(m * 100 + col)
+                got = out[lane, f, j].item()
+                assert got == expect, (
+                    f"lane={lane} f={f} j={j}: got {got}, want {expect}")
+
+
 @pytest.mark.parametrize("ks,C,K", [(7, 12, 128), (5, 128, 128), (3, 32, 3),
                                     (1, 128, 64)])
 def test_conv_backward_parity(ext, ks, C, K):

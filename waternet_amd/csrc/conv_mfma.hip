@@ -70,8 +70,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 * BM*32
-  bf16_t* lB = lA + 2 * BM * 32;                             // 2 * BN*32
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // BM*32
+  bf16_t* lB = lA + BM * 32;                                 // BN*32
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -187,16 +187,14 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     }
   };
 
-  auto writeTiles = [&](int buf) {
+  auto writeTiles = [&]() {
 #pragma unroll
     for (int s = 0; s < 2; ++s)
-      *reinterpret_cast<bf16x8*>(lA + buf * (BM * 32) + (tid + s * 256) * 8) =
-          aR[s];
+      *reinterpret_cast<bf16x8*>(lA + (tid + s * 256) * 8) = aR[s];
 #pragma unroll
     for (int s = 0; s < NBS; ++s)
       if (bV[s])
-        *reinterpret_cast<bf16x8*>(lB + buf * (BN * 32) + (tid + s * 256) * 8) =
-            bR[s];
+        *reinterpret_cast<bf16x8*>(lB + (tid + s * 256) * 8) = bR[s];
   };
 
   f32x4 acc[FM][FN];
@@ -216,14 +214,17 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   }
   loadA(ks0 * 32);
   loadB(ks0 * 32);
-  writeTiles(0);
+  writeTiles();
   __syncthreads();
 
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
+  // Single-buffer pipeline: per k-step, prefetch the next tiles from global
+  // into registers, pull this step's fragments out of LDS, barrier, then
+  // restage LDS for the next step back-to-back with this step's MFMAs (the
+  // ds_writes are independent of the matrix math, so they hide under it).
   for (int ks = ks0; ks < nk; ++ks) {
-    const int cur = (ks - ks0) & 1;
     if (ks + 1 < nk) {
       loadA((ks + 1) * 32);
       loadB((ks + 1) * 32);
@@ -233,25 +234,23 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     for (int fm = 0; fm < FM; ++fm) {
       int mfG = wr * FM + fm;
       aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
+          lA + ((mfG * 4 + lg) * 16 + li) * 8);
     }
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
       int nfG = wc * FN + fn;
       bF[fn] = *reinterpret_cast<const bf16x8*>(
-          lB + cur * (BN * 32) + ((nfG * 4 + lg) * 16 + li) * 8);
+          lB + ((nfG * 4 + lg) * 16 + li) * 8);
     }
+    __syncthreads();  // all fragments of step ks are in registers
+    if (ks + 1 < nk) writeTiles();
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
       for (int fn = 0; fn < FN; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    if (ks + 1 < nk) {
-      __syncthreads();
-      writeTiles(cur ^ 1);
-      __syncthreads();
-    }
+    if (ks + 1 < nk) __syncthreads();  // restage visible before next reads
   }
 
   // ---- epilogue: bias + activation (+ pad-channel zeroing), or fp32
@@ -303,22 +302,50 @@ __global__ void k_splitk_finalize(const float* __restrict__ Y32,
 // ---------------------------------------------------------------------------
 // Weight gradient: dW[k][rsc] = sum_m dY[m][k] * X[m][rsc]
 //
-// v2 (tuned from rocprofv3 r01 stats where wgrad was 33% of step time):
+// v3: both GEMM operands are [reduction=m]-major in global memory (dY is
+// (m,k), X gathers are (m,c)), i.e. TRANSPOSED relative to the MFMA
+// fragment layout (lane needs 8 m for a fixed column). v2 transposed via
+// 8x ds_write_b16 scatter, which lands 16-32 lanes on the same mod-32
+// write bank (ds_write banks are (a/4)%32) — LDS-write-bound. v3 stores
+// the tiles m-major with vector ds_write_b128 (conflict-free by kblk
+// stride choice) and uses gfx950's ds_read_b64_tr_b16 hardware transpose
+// read to deliver MFMA fragments (guide T10, the attention-V recipe).
+//
+// LDS image per operand: [col/16 kblk][m/4 mblk][4m][16col] subtiles,
+// row-major inside a subtile; mblk stride 96 elems (2*96 dwords = 32 mod
+// 64 so tr-read lane groups land on disjoint read banks), kblk stride
+// 1552 elems (776 dwords = 8 mod 32 so the staging b128 writes of one
+// 8-lane service group hit 8 distinct bank quads).
+//
 //   - BK x 128 output tile per block (BK = min(Kp,128)), 2x2 / 1x4 waves
-//   - m-chunks of 64 (two MFMA k-steps); single LDS buffer with hoisted
-//     fragment reads: per chunk, all ds_read_b128 fragments for the chunk
-//     are pulled into registers first, then after one barrier the staging
-//     writes for the NEXT chunk issue back-to-back with the chunk's MFMAs
-//     (independent, so the LDS writes hide under matrix-core work).
-//   - LDS is row-major [row][m] with +8 element row padding: staging writes
-//     come from contiguous-k global vector loads (bf16x8) and scatter 8
-//     ds_write_b16; fragment reads are aligned ds_read_b128 with 36-dword
-//     row stride (16 distinct banks, conflict-free).
+//   - m-chunks of 64 (two MFMA k-steps); single LDS buffer: tr-read all
+//     fragments into registers, barrier, restage next chunk's tiles
+//     back-to-back with this chunk's MFMAs, barrier.
 //   - split-m z dimension sized to fill 256 CUs x 2 blocks; gy (rsc tiles)
 //     varies fastest so co-resident blocks share dY/X chunks through L2.
 //   - fp32 atomicAdd epilogue into the NCHW fp32 grad tensor (pre-zeroed).
 // Requires the KMAP-0 fragment layout (frag_k16(g,e) = g*8+e).
 // ---------------------------------------------------------------------------
+
+constexpr int TR_MBS = 96;    // mblk stride (elements)
+constexpr int TR_KBS = 1552;  // kblk stride (elements)
+
+// elem offset of (m, col) inside a tr image
+WN_DEVFN int tr_addr(int m, int col) {
+  return (col >> 4) * TR_KBS + (m >> 2) * TR_MBS + (m & 3) * 16 + (col & 15);
+}
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// ds_read_b64_tr_b16: lane (l&15) of each 16-lane group receives column
+// (l&15), rows 0..3, of the [4][16] row-major bf16 subtile whose 16
+// 8-byte pieces the group's lanes address in canonical order (lane i ->
+// subtile_base + i*8B). addr is an LDS byte address.
+WN_DEVFN bf16x4 ds_tr16(unsigned addr) {
+  bf16x4 v;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(addr));
+  return v;
+}
 
 template <int KS, int BK>
 __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
@@ -332,7 +359,6 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   constexpr int RS = KS * KS;
   constexpr int BR = 128;           // rsc tile
   constexpr int CH = 64;            // m chunk
-  constexpr int LROW = CH + 8;      // padded LDS row (elements)
   constexpr int WR = (BK >= 64) ? 2 : 1;   // wave rows (k dim)
   constexpr int WC = 4 / WR;               // wave cols (rsc dim)
   constexpr int FK = BK / WR / 16;  // k fragments per wave
@@ -345,8 +371,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const int HW = H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // BK x LROW
-  bf16_t* lB = lA + BK * LROW;                    // BR x LROW
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // (BK/16) x TR_KBS
+  bf16_t* lB = lA + (BK / 16) * TR_KBS;           // (BR/16) x TR_KBS
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -420,17 +446,13 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 #pragma unroll
     for (int s = 0; s < ASLOT; ++s) {
       if (!aV[s]) continue;
-      bf16_t* dst = lA + aJg[s] * 8 * LROW + aMl[s];
-#pragma unroll
-      for (int e = 0; e < 8; ++e) dst[e * LROW] = aR[s][e];
+      *reinterpret_cast<bf16x8*>(lA + tr_addr(aMl[s], aJg[s] * 8)) = aR[s];
     }
 #pragma unroll
     for (int s = 0; s < BSLOT; ++s) {
       int slot = tid + s * 256;
       int jg = slot % (BR / 8);
-      bf16_t* dst = lB + jg * 8 * LROW + bMl[s];
-#pragma unroll
-      for (int e = 0; e < 8; ++e) dst[e * LROW] = bR[s][e];
+      *reinterpret_cast<bf16x8*>(lB + tr_addr(bMl[s], jg * 8)) = bR[s];
     }
   };
 
@@ -442,6 +464,12 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 
   const long nChunks = (M + CH - 1) / CH;
   const int lg = lane >> 4, li = lane & 15;
+  // per-lane tr-read byte addresses for kstep 0 / mblk (2*lg); the second
+  // 4-m half (e=4..7) is the next mblk at +2*TR_MBS bytes.
+  const unsigned aTr0 = (unsigned)(unsigned long long)lA +
+                        (unsigned)(2 * lg * TR_MBS + li * 4) * 2;
+  const unsigned bTr0 = (unsigned)(unsigned long long)lB +
+                        (unsigned)(2 * lg * TR_MBS + li * 4) * 2;
 
   long chunk = blockIdx.z;
   if (chunk < nChunks) {
@@ -456,29 +484,42 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
       loadA(next * CH);  // global -> regs, overlaps this chunk's math
       loadB(next * CH);
     }
-    // hoist ALL fragment reads for this chunk into registers
-    bf16x8 aF[CH / 32][FK], bF[CH / 32][FR];
+    // tr-read ALL fragments for this chunk into registers (guide T10)
+    bf16x4 aT[CH / 32][FK][2], bT[CH / 32][FR][2];
 #pragma unroll
-    for (int ks32 = 0; ks32 < CH / 32; ++ks32) {
+    for (int s2 = 0; s2 < CH / 32; ++s2) {
 #pragma unroll
-      for (int f = 0; f < FK; ++f)
-        aF[ks32][f] = *reinterpret_cast<const bf16x8*>(
-            lA + ((wr * FK + f) * 16 + li) * LROW + ks32 * 32 + lg * 8);
+      for (int f = 0; f < FK; ++f) {
+        const unsigned base = aTr0 + (unsigned)((wr * FK + f) * TR_KBS +
+                                                s2 * 8 * TR_MBS) * 2;
+        aT[s2][f][0] = ds_tr16(base);
+        aT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+      }
 #pragma unroll
-      for (int f = 0; f < FR; ++f)
-        bF[ks32][f] = *reinterpret_cast<const bf16x8*>(
-            lB + ((wc * FR + f) * 16 + li) * LROW + ks32 * 32 + lg * 8);
+      for (int f = 0; f < FR; ++f) {
+        const unsigned base = bTr0 + (unsigned)((wc * FR + f) * TR_KBS +
+                                                s2 * 8 * TR_MBS) * 2;
+        bT[s2][f][0] = ds_tr16(base);
+        bT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+      }
     }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);  // keep MFMAs below the wait
     __syncthreads();  // all waves done reading; LDS free for restaging
     if (next < nChunks) writeTiles();  // issues alongside the MFMAs below
 #pragma unroll
-    for (int ks32 = 0; ks32 < CH / 32; ++ks32)
+    for (int s2 = 0; s2 < CH / 32; ++s2)
 #pragma unroll
       for (int fa = 0; fa < FK; ++fa)
 #pragma unroll
-        for (int fb = 0; fb < FR; ++fb)
+        for (int fb = 0; fb < FR; ++fb) {
+          const bf16x8 av = __builtin_shufflevector(
+              aT[s2][fa][0], aT[s2][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
+          const bf16x8 bv = __builtin_shufflevector(
+              bT[s2][fb][0], bT[s2][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
           acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aF[ks32][fa], bF[ks32][fb], acc[fa][fb], 0, 0, 0);
+              av, bv, acc[fa][fb], 0, 0, 0);
+        }
     if (next < nChunks) __syncthreads();  // writes visible before next reads
   }
 
@@ -505,34 +546,91 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 }
 
 // ---------------------------------------------------------------------------
+// tr-read semantics probe: one wave fills a tr image with addr-coded values
+// via the SAME writeTiles addressing (tr_addr), tr-reads fragments the same
+// way wgrad does, and writes what each (lane, frag, elem) received. The
+// host test checks element (lane g*16+i, reg j) == value coded (m=..,col=..).
+// ---------------------------------------------------------------------------
+
+__global__ void k_probe_tr(float* __restrict__ out /* [64][2][4] */) {
+  __shared__ __attribute__((aligned(16))) bf16_t img[2 * TR_KBS];
+  const int lane = threadIdx.x & 63;
+  // fill 64 m x 32 col with value m*100 + col (bf16-exact for m<..,col<32)
+  for (int m = lane; m < 64; m += 64)
+    for (int col = 0; col < 32; ++col)
+      img[tr_addr(m, col)] = f2bf((float)(m * 100 + col));
+  __syncthreads();
+  const int lg = lane >> 4, li = lane & 15;
+  const unsigned base0 = (unsigned)(unsigned long long)img +
+                         (unsigned)(2 * lg * TR_MBS + li * 4) * 2;
+#pragma unroll
+  for (int f = 0; f < 2; ++f) {  // two col-blocks (kblk 0, 1)
+    const unsigned b = base0 + (unsigned)(f * TR_KBS) * 2;
+    bf16x4 lo = ds_tr16(b);
+    bf16x4 hi = ds_tr16(b + TR_MBS * 2);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[(lane * 2 + f) * 8 + j] = bf2f(lo[j]);
+      out[(lane * 2 + f) * 8 + 4 + j] = bf2f(hi[j]);
+    }
+  }
+}
+
+at::Tensor probe_tr() {
+  auto out = at::zeros({64, 2, 8}, at::TensorOptions()
+                                       .dtype(at::kFloat)
+                                       .device(at::kCUDA));
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_probe_tr, dim3(1), dim3(64), 0, stream,
+                     out.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // Bias gradient: db[k] = sum_m dY[m][k]
 // ---------------------------------------------------------------------------
 
+// Each thread owns one 8-wide k-group and vector-loads bf16x8 rows of dY,
+// accumulating 8 fp32 partials in registers; threads covering the same
+// k-group at different m-phases reduce through LDS, then one atomicAdd per
+// (block, k). Grid: (ceil(Kp/8/KGR), msplit) with KGR k-groups per block.
 __global__ void k_bias_grad(const bf16_t* __restrict__ dY,
                             float* __restrict__ dB, long M, int Kp, int K,
                             int msplit) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* red = reinterpret_cast<float*>(smem);
-  const int KX = min(Kp, 64);
-  const int ROWS = 256 / KX;
-  const int kx = threadIdx.x % KX;
-  const int my = threadIdx.x / KX;
-  const int kbase = blockIdx.x * KX;
-  const int k = kbase + kx;
+  float* red = reinterpret_cast<float*>(smem);  // [256][8] fp32
+  const int ngrp = Kp >> 3;                  // 8-wide k-groups in Kp
+  const int KGR = min(ngrp, 32);             // k-groups per block
+  const int MR = 256 / KGR;                  // m-rows strided per block
+  const int kg = threadIdx.x % KGR;          // group id (fastest: coalesced)
+  const int mr = threadIdx.x / KGR;
+  const int k0 = (blockIdx.x * KGR + kg) * 8;
 
-  float s = 0.f;
-  const long rows_per_split = (M + msplit - 1) / msplit;
-  const long mstart = (long)blockIdx.y * rows_per_split;
-  const long mend = min(mstart + rows_per_split, M);
-  if (k < Kp)
-    for (long m = mstart + my; m < mend; m += ROWS)
-      s += bf2f(dY[m * Kp + k]);
-  red[my * KX + kx] = s;
+  float acc[8] = {};
+  const long rows = (M + msplit - 1) / msplit;
+  const long mstart = (long)blockIdx.y * rows;
+  const long mend = min(mstart + rows, M);
+  if (k0 < Kp) {
+    for (long m = mstart + mr; m < mend; m += MR) {
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(dY + m * Kp + k0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += bf2f(v[e]);
+    }
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = acc[e];
   __syncthreads();
-  if (my == 0 && k < K) {
-    float t = 0.f;
-    for (int r = 0; r < ROWS; ++r) t += red[r * KX + kx];
-    atomicAdd(&dB[k], t);
+  // tree-reduce over the MR rows that share this thread's k-group
+  if (mr == 0 && k0 < Kp) {
+#pragma unroll 1
+    for (int r = 1; r < MR; ++r)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += red[(r * KGR + kg) * 8 + e];
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      if (k0 + e < K) atomicAdd(&dB[k0 + e], acc[e]);
   }
 }
 
@@ -602,7 +700,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
   auto launch = [&](auto bn_const) {
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
-    const size_t lds = (2 * 128 * 32 + 2 * BN * 32) * sizeof(bf16_t);
+    const size_t lds = (128 * 32 + BN * 32) * sizeof(bf16_t);
     // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
     // 256 CUs idle; split the K loop across gz slices into fp32 partials,
     // then finalize bias+act+bf16 in a second tiny pass.
@@ -690,7 +788,7 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   const long nChunks = ((long)N * H * W + 63) / 64;
   split = (int)std::min<long>(split, nChunks);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
-  const size_t lds = (size_t)(BK + 128) * 72 * sizeof(bf16_t);
+  const size_t lds = (size_t)(BK / 16 + 128 / 16) * TR_KBS * sizeof(bf16_t);
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
@@ -725,13 +823,16 @@ void bias_grad(const at::Tensor& dy, at::Tensor& db) {
   const int Kp = dy.size(3);
   const long M = dy.numel() / Kp;
   const int K = db.size(0);
-  const int KX = std::min(Kp, 64);
-  const int gx = (Kp + KX - 1) / KX;
-  int msplit = (int)std::min<long>(64, (M + 255) / 256);
-  msplit = std::max(msplit, 1);
+  const int ngrp = Kp / 8;
+  const int KGR = std::min(ngrp, 32);
+  const int gx = (ngrp + KGR - 1) / KGR;
+  // fill the chip: ~512 blocks, each covering M/msplit rows
+  int msplit = std::max(1, 512 / gx);
+  const int MR = 256 / KGR;
+  msplit = (int)std::min<long>(msplit, (M + MR - 1) / MR);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(k_bias_grad, dim3(gx, msplit), dim3(256),
-                     256 * sizeof(float), stream,
+                     256 * 8 * sizeof(float), stream,
                      (const bf16_t*)dy.data_ptr(), db.data_ptr<float>(), M,
                      Kp, K, msplit);
   HIP_CHECK_LAST();
