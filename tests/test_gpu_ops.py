@@ -88,7 +88,8 @@ def test_tr_read_lane_mapping(ext):
             for j in range(8):
                 m = 8 * g + j if j < 4 else 8 * g + (j - 4) + 4
                 col = f * 16 + i
-                expect = float(m * 100 + col)
+                expect = torch.tensor(
+                    float(m * 100 + col)).bfloat16().float().item()
                 got = out[lane, f, j].item()
                 assert got == expect, (
                     f"lane={lane} f={f} j={j}: got {got}, want {expect}")

@@ -13,6 +13,7 @@ void bias_grad(const at::Tensor& dy, at::Tensor& db);
 at::Tensor pack_weight_fwd(const at::Tensor& w, int64_t Kp, int64_t Cp);
 at::Tensor pack_weight_dgrad(const at::Tensor& w, int64_t Kp, int64_t Cp);
 at::Tensor probe_tr();
+at::Tensor probe_tr_raw();
 
 // elementwise.hip
 std::vector<at::Tensor> build_inputs(const at::Tensor& raw,
@@ -61,6 +62,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_weight_fwd", &pack_weight_fwd);
   m.def("pack_weight_dgrad", &pack_weight_dgrad);
   m.def("probe_tr", &probe_tr, "ds_read_b64_tr_b16 lane-mapping probe");
+  m.def("probe_tr_raw", &probe_tr_raw, "raw tr permutation dump");
   m.def("build_inputs", &build_inputs);
   m.def("nchw_to_nhwc", &nchw_to_nhwc);
   m.def("nhwc_to_nchw", &nhwc_to_nchw);

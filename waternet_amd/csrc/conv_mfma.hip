@@ -343,7 +343,9 @@ typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 // subtile_base + i*8B). addr is an LDS byte address.
 WN_DEVFN bf16x4 ds_tr16(unsigned addr) {
   bf16x4 v;
-  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(addr));
+  // "memory" keeps the read below the staging stores / barrier (without it
+  // hipcc hoists the first tr read above the LDS fill — measured garbage).
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(addr) : "memory");
   return v;
 }
 
@@ -577,12 +579,39 @@ __global__ void k_probe_tr(float* __restrict__ out /* [64][2][4] */) {
   }
 }
 
+// Raw semantics probe: linear image img[e] = e; lane l reads 8 B at
+// byte offset l*8 (elements 4l..4l+3). out[l][j] = delivered element
+// index — exposes the hardware's lane/element permutation directly.
+__global__ void k_probe_tr_raw(float* __restrict__ out /* [64][4] */) {
+  __shared__ __attribute__((aligned(16))) bf16_t img[256];
+  const int lane = threadIdx.x & 63;
+  for (int e = lane; e < 256; e += 64) img[e] = f2bf((float)e);
+  __syncthreads();
+  const unsigned addr =
+      (unsigned)(unsigned long long)img + (unsigned)lane * 8;
+  bf16x4 v = ds_tr16(addr);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = bf2f(v[j]);
+}
+
 at::Tensor probe_tr() {
   auto out = at::zeros({64, 2, 8}, at::TensorOptions()
                                        .dtype(at::kFloat)
                                        .device(at::kCUDA));
   hipStream_t stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(k_probe_tr, dim3(1), dim3(64), 0, stream,
+                     out.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return out;
+}
+
+at::Tensor probe_tr_raw() {
+  auto out = at::zeros({64, 4}, at::TensorOptions()
+                                    .dtype(at::kFloat)
+                                    .device(at::kCUDA));
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_probe_tr_raw, dim3(1), dim3(64), 0, stream,
                      out.data_ptr<float>());
   HIP_CHECK_LAST();
   return out;
@@ -826,8 +855,10 @@ void bias_grad(const at::Tensor& dy, at::Tensor& db) {
   const int ngrp = Kp / 8;
   const int KGR = std::min(ngrp, 32);
   const int gx = (ngrp + KGR - 1) / KGR;
-  // fill the chip: ~512 blocks, each covering M/msplit rows
-  int msplit = std::max(1, 512 / gx);
+  // Modest split: dB has only K<=512 fp32 entries (a few cache lines), so
+  // msplit*K atomics serialize on them — 64 blocks saturate HBM read bw
+  // for these sizes while keeping atomic pressure ~64 per address.
+  int msplit = std::max(1, 64 / gx);
   const int MR = 256 / KGR;
   msplit = (int)std::min<long>(msplit, (M + MR - 1) / MR);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
