@@ -56,6 +56,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     const float* __restrict__ Bias, // [>=Klog] or nullptr
     bf16_t* __restrict__ Y,         // (N, H, W, Kp)
     int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act,
+    const bf16_t* __restrict__ Zero16,    // 16 B of zeros (halo/pad source)
     float* __restrict__ Y32 = nullptr) {  // SPLITK: fp32 partials (pre-zeroed)
   constexpr int BM = 128;
   constexpr int PAD = KS / 2;
@@ -66,12 +67,14 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   constexpr int FN = BN / WN / 16;  // fragments per wave in N
   constexpr int SLOTS_B = BN * 4;
   constexpr int NBS = (SLOTS_B + 255) / 256;  // B slots per thread (1 or 2)
+  constexpr int LB = NBS * 256 * 8;  // B buffer elems (>= SLOTS_B*8, so the
+                                     // glds of invalid slots lands in-pad)
   const int KG = RS * Cp;
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // BM*32
-  bf16_t* lB = lA + BM * 32;                                 // BN*32
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 x BM*32
+  bf16_t* lB = lA + 2 * BM * 32;                             // 2 x LB
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -114,87 +117,38 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     bKb[s] = kb;
   }
 
-  bf16x8 aR[2], bR[NBS];
+  static_assert(WN_MFMA_KMAP == 0, "glds staging assumes KMAP 0");
 
-  auto loadA = [&](int k0) {
+  // Stage one K-step's A and B tiles straight into LDS buffer `buf` with
+  // global_load_lds (16 B per lane, lane-linear LDS image — guide §5
+  // "common mistake 1": width-16 glds is the staging lever). Out-of-range
+  // lanes (conv halo, channel pad, M tail) read from Zero16 instead —
+  // every lane always issues its DMA so the image is fully defined.
+  auto stage = [&](int buf, int k0) {
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
-      bf16x8 v = {};
-      if (aMv[s]) {
-#if WN_MFMA_KMAP == 0
-        // 8 contiguous k: one 16B load (within one tap: Cp % 8 == 0)
-        int rsc = k0 + aKb[s] * 8;
-        if (rsc < KG) {
-          int tap = rsc >> log2Cp;
-          int c = rsc & (Cp - 1);
-          int dy = tap / KS, dx = tap - (tap / KS) * KS;
-          int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
-          if (iy >= 0 && iy < H && ix >= 0 && ix < W)
-            v = *reinterpret_cast<const bf16x8*>(
-                X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c);
-        }
-#else
-        // two 4-half pieces (k = 4g.. and k = 16+4g..)
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-          int rsc = k0 + aKb[s] * 4 + h * 16;
-          if (rsc < KG) {
-            int tap = rsc >> log2Cp;
-            int c = rsc & (Cp - 1);
-            int dy = tap / KS, dx = tap - (tap / KS) * KS;
-            int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
-            if (iy >= 0 && iy < H && ix >= 0 && ix < W) {
-              const ushort2 p = *reinterpret_cast<const ushort2*>(
-                  X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c);
-              const ushort2 q = *reinterpret_cast<const ushort2*>(
-                  X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c + 2);
-              reinterpret_cast<ushort2*>(&v)[h * 2] = p;
-              reinterpret_cast<ushort2*>(&v)[h * 2 + 1] = q;
-            }
-          }
-        }
-#endif
+      const bf16_t* src = Zero16;
+      int rsc = k0 + aKb[s] * 8;
+      if (aMv[s] && rsc < KG) {
+        int tap = rsc >> log2Cp;
+        int c = rsc & (Cp - 1);
+        int dy = tap / KS, dx = tap - (tap / KS) * KS;
+        int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
+        if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+          src = X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c;
       }
-      aR[s] = v;
+      __builtin_amdgcn_global_load_lds(
+          src, lA + buf * (BM * 32) + (tid + s * 256) * 8, 16, 0, 0);
     }
-  };
-
-  auto loadB = [&](int k0) {
 #pragma unroll
     for (int s = 0; s < NBS; ++s) {
-      bf16x8 v = {};
-      if (bV[s] && bK[s] < Kp) {
-#if WN_MFMA_KMAP == 0
-        int rsc = k0 + bKb[s] * 8;
-        if (rsc < KG)
-          v = *reinterpret_cast<const bf16x8*>(Wp + (long)bK[s] * KG + rsc);
-#else
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-          int rsc = k0 + bKb[s] * 4 + h * 16;
-          if (rsc < KG) {
-            const ushort2 p = *reinterpret_cast<const ushort2*>(
-                Wp + (long)bK[s] * KG + rsc);
-            const ushort2 q = *reinterpret_cast<const ushort2*>(
-                Wp + (long)bK[s] * KG + rsc + 2);
-            reinterpret_cast<ushort2*>(&v)[h * 2] = p;
-            reinterpret_cast<ushort2*>(&v)[h * 2 + 1] = q;
-          }
-        }
-#endif
-      }
-      bR[s] = v;
+      const bf16_t* src = Zero16;
+      int rsc = k0 + bKb[s] * 8;
+      if (bV[s] && bK[s] < Kp && rsc < KG)
+        src = Wp + (long)bK[s] * KG + rsc;
+      __builtin_amdgcn_global_load_lds(
+          src, lB + buf * LB + (tid + s * 256) * 8, 16, 0, 0);
     }
-  };
-
-  auto writeTiles = [&]() {
-#pragma unroll
-    for (int s = 0; s < 2; ++s)
-      *reinterpret_cast<bf16x8*>(lA + (tid + s * 256) * 8) = aR[s];
-#pragma unroll
-    for (int s = 0; s < NBS; ++s)
-      if (bV[s])
-        *reinterpret_cast<bf16x8*>(lB + (tid + s * 256) * 8) = bR[s];
   };
 
   f32x4 acc[FM][FN];
@@ -212,45 +166,39 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     nk = min(nkAll, ks0 + per);
     if (ks0 >= nk) return;
   }
-  loadA(ks0 * 32);
-  loadB(ks0 * 32);
-  writeTiles();
-  __syncthreads();
+  stage(0, ks0 * 32);
+  __syncthreads();  // drains the glds (vmcnt 0) + barrier
 
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
-  // Single-buffer pipeline: per k-step, prefetch the next tiles from global
-  // into registers, pull this step's fragments out of LDS, barrier, then
-  // restage LDS for the next step back-to-back with this step's MFMAs (the
-  // ds_writes are independent of the matrix math, so they hide under it).
+  // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next tile's
+  // DMA first, ds_read + MFMA the current buffer, one vmcnt(0)+barrier per
+  // K-step (the __syncthreads drains the in-flight DMA).
+  int cur = 0;
   for (int ks = ks0; ks < nk; ++ks) {
-    if (ks + 1 < nk) {
-      loadA((ks + 1) * 32);
-      loadB((ks + 1) * 32);
-    }
+    if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
     bf16x8 aF[FM], bF[FN];
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm) {
       int mfG = wr * FM + fm;
       aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + ((mfG * 4 + lg) * 16 + li) * 8);
+          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
     }
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
       int nfG = wc * FN + fn;
       bF[fn] = *reinterpret_cast<const bf16x8*>(
-          lB + ((nfG * 4 + lg) * 16 + li) * 8);
+          lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
     }
-    __syncthreads();  // all fragments of step ks are in registers
-    if (ks + 1 < nk) writeTiles();
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
 #pragma unroll
       for (int fn = 0; fn < FN; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    if (ks + 1 < nk) __syncthreads();  // restage visible before next reads
+    if (ks + 1 < nk) __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: bias + activation (+ pad-channel zeroing), or fp32
@@ -725,11 +673,17 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
       bias.has_value() ? bias->data_ptr<float>() : nullptr;
   const int gx = (int)((M + 127) / 128);
   const int nk = (KS * KS * Cp + 31) / 32;
+  // 16 B zero source for glds halo/pad lanes (per-device, created once)
+  static thread_local at::Tensor zero16;
+  if (!zero16.defined() || zero16.device() != x.device())
+    zero16 = at::zeros({8}, x.options());
+  const bf16_t* zptr = (const bf16_t*)zero16.data_ptr();
 
   auto launch = [&](auto bn_const) {
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
-    const size_t lds = (128 * 32 + BN * 32) * sizeof(bf16_t);
+    constexpr int NBS = (BN * 4 + 255) / 256;
+    const size_t lds = (2 * 128 * 32 + 2 * NBS * 256 * 8) * sizeof(bf16_t);
     // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
     // 256 CUs idle; split the K loop across gz slices into fp32 partials,
     // then finalize bias+act+bf16 in a second tiny pass.
@@ -743,7 +697,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
                          (const bf16_t*)x.data_ptr(),
                          (const bf16_t*)wp.data_ptr(), bptr,
                          (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
-                         Klog, act, y32.data_ptr<float>());
+                         Klog, act, zptr, y32.data_ptr<float>());
       const long total = M * Kp;
       const int fb = (int)std::min<long>(1024, (total + 255) / 256);
       hipLaunchKernelGGL(k_splitk_finalize, dim3(fb), dim3(256), 0, stream,
@@ -754,7 +708,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
                          lds, stream, (const bf16_t*)x.data_ptr(),
                          (const bf16_t*)wp.data_ptr(), bptr,
                          (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
-                         Klog, act);
+                         Klog, act, zptr);
     }
   };
   if (Kp >= 128)
