@@ -78,21 +78,18 @@ def test_conv_fwd_sigmoid(ext):
 
 
 def test_tr_read_lane_mapping(ext):
-    """ds_read_b64_tr_b16 must deliver column (lane&15), rows mblk*4..+3 of
-    the [4][16] subtile addressed by the wgrad tr image layout — the wgrad
-    kernel's fragment correctness rests on exactly this mapping."""
-    out = ext.probe_tr().cpu()  # (64, 2, 8): lane, col-block, elem
+    """ds_read_b64_tr_b16 semantics pin: with canonical per-lane addresses
+    (lane l reads 8 B at offset l*8 of a linear image), lane l must receive
+    elements {base + (l&15) + 16j} — column (l&15), rows j, of the [4][16]
+    row-major block its 16-lane group stages (guide T10). The wgrad kernel's
+    fragment layout rests on exactly this mapping."""
+    out = ext.probe_tr_raw().cpu()  # (64, 4)
     for lane in range(64):
         g, i = lane >> 4, lane & 15
-        for f in range(2):
-            for j in range(8):
-                m = 8 * g + j if j < 4 else 8 * g + (j - 4) + 4
-                col = f * 16 + i
-                expect = torch.tensor(
-                    float(m * 100 + col)).bfloat16().float().item()
-                got = out[lane, f, j].item()
-                assert got == expect, (
-                    f"lane={lane} f={f} j={j}: got {got}, want {expect}")
+        for j in range(4):
+            expect = float(g * 64 + i + 16 * j)
+            got = out[lane, j].item()
+            assert got == expect, f"lane={lane} j={j}: got {got}, want {expect}"
 
 
 @pytest.mark.parametrize("ks,C,K", [(7, 12, 128), (5, 128, 128), (3, 32, 3),

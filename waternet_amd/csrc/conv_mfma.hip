@@ -158,13 +158,13 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nkAll = (KG + 31) / 32;
-  // SPLITK: z-slice [ks0, nk) of the K loop; else the whole range.
+  // SPLITK: z-slice [ks0, nk) of the K loop; else the whole range. Floor
+  // slicing keeps every slice non-empty for gz <= nkAll, so every fp32
+  // slab is fully written (the finalize pass sums all of them).
   int ks0 = 0, nk = nkAll;
   if (SPLITK) {
-    const int per = (nkAll + gridDim.z - 1) / gridDim.z;
-    ks0 = blockIdx.z * per;
-    nk = min(nkAll, ks0 + per);
-    if (ks0 >= nk) return;
+    ks0 = (int)((long)blockIdx.z * nkAll / gridDim.z);
+    nk = (int)((long)(blockIdx.z + 1) * nkAll / gridDim.z);
   }
   stage(0, ks0 * 32);
   __syncthreads();  // drains the glds (vmcnt 0) + barrier
@@ -217,7 +217,9 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
         long m = m0 + (wr * FM + fm) * 16 + lr4 + r;
         if (m >= M) continue;
         if (SPLITK) {
-          atomicAdd(&Y32[m * Kp + k], acc[fm][fn][r]);
+          // per-slice fp32 slab, plain stores (no atomics — the finalize
+          // pass reduces over gridDim.z slabs)
+          Y32[((long)blockIdx.z * M + m) * Kp + k] = acc[fm][fn][r];
         } else {
           float v = acc[fm][fn][r] + bv;
           if (act == ACT_RELU) v = fmaxf(v, 0.f);
@@ -230,20 +232,30 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   }
 }
 
-// Split-K finalize: Y = act(Y32 + bias), pad channels zeroed.
+// Split-K finalize: Y = act(sum over gz slabs of Y32 + bias), pad zeroed.
 __global__ void k_splitk_finalize(const float* __restrict__ Y32,
                                   const float* __restrict__ Bias,
                                   bf16_t* __restrict__ Y, long M, int Kp,
-                                  int Klog, int act) {
+                                  int Klog, int act, int gz) {
   const long total = M * Kp;
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
-    const int k = (int)(i % Kp);
-    float v = Y32[i] + ((Bias != nullptr && k < Klog) ? Bias[k] : 0.f);
-    if (act == ACT_RELU) v = fmaxf(v, 0.f);
-    else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
-    if (k >= Klog) v = 0.f;
-    Y[i] = f2bf(v);
+  for (long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i0 < total; i0 += (long)gridDim.x * blockDim.x * 4) {
+    f32x4 v = *reinterpret_cast<const f32x4*>(Y32 + i0);
+    for (int z = 1; z < gz; ++z) {
+      const f32x4 s = *reinterpret_cast<const f32x4*>(Y32 + z * total + i0);
+#pragma unroll
+      for (int e = 0; e < 4; ++e) v[e] += s[e];
+    }
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const long i = i0 + e;
+      const int k = (int)(i % Kp);
+      float w = v[e] + ((Bias != nullptr && k < Klog) ? Bias[k] : 0.f);
+      if (act == ACT_RELU) w = fmaxf(w, 0.f);
+      else if (act == ACT_SIGMOID) w = 1.f / (1.f + __expf(-w));
+      if (k >= Klog) w = 0.f;
+      Y[i] = f2bf(w);
+    }
   }
 }
 
@@ -691,7 +703,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     int gz = 1;
     if (base < 320 && nk >= 16) gz = std::min(std::max(1, 512 / base), nk / 4);
     if (gz > 1) {
-      auto y32 = at::zeros({M, (long)Kp}, x.options().dtype(at::kFloat));
+      auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
       hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
                          dim3(256), lds, stream,
                          (const bf16_t*)x.data_ptr(),
@@ -699,10 +711,10 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
                          (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
                          Klog, act, zptr, y32.data_ptr<float>());
       const long total = M * Kp;
-      const int fb = (int)std::min<long>(1024, (total + 255) / 256);
+      const int fb = (int)std::min<long>(512, (total / 4 + 255) / 256);
       hipLaunchKernelGGL(k_splitk_finalize, dim3(fb), dim3(256), 0, stream,
                          y32.data_ptr<float>(), bptr,
-                         (bf16_t*)y.data_ptr(), M, Kp, Klog, act);
+                         (bf16_t*)y.data_ptr(), M, Kp, Klog, act, gz);
     } else {
       hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256),
                          lds, stream, (const bf16_t*)x.data_ptr(),

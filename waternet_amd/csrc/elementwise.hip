@@ -63,29 +63,61 @@ __global__ void k_build_inputs(const float* __restrict__ raw,
 // NCHW fp32 <-> NHWC bf16 bridges
 // ---------------------------------------------------------------------------
 
+// LDS-tiled layout bridges: 256 pixels x 16 channels per block so BOTH
+// sides are coalesced (the naive per-pixel loop read NCHW at stride HW and
+// wrote 2 B scattered — measured 125 GB/s; these run at HBM rate).
 __global__ void k_nchw2nhwc(const float* __restrict__ x,
                             bf16_t* __restrict__ y, long NHW, long HW, int C,
                             int Cp) {
-  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
-       p += (long)gridDim.x * blockDim.x) {
-    const long n = p / HW, rem = p - n * HW;
-    const float* src = x + (n * C) * HW + rem;
-    bf16_t* dst = y + p * Cp;
-    int c = 0;
-    for (; c < C; ++c) dst[c] = f2bf(src[(long)c * HW]);
-    for (; c < Cp; ++c) dst[c] = f2bf(0.f);
+  __shared__ float tile[16][257];
+  const int tid = threadIdx.x;
+  const long p = (long)blockIdx.x * 256 + tid;
+  const int cc0 = blockIdx.z * 16;
+  const bool pv = p < NHW;
+  const long n = pv ? p / HW : 0;
+  const long rem = p - n * HW;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const int c = cc0 + i;
+    tile[i][tid] =
+        (pv && c < C) ? x[(n * C + c) * HW + rem] : 0.f;
+  }
+  __syncthreads();
+  if (pv) {
+#pragma unroll
+    for (int v8 = 0; v8 < 2; ++v8) {
+      bf16x8 v;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v[e] = f2bf(tile[v8 * 8 + e][tid]);
+      *reinterpret_cast<bf16x8*>(y + p * Cp + cc0 + v8 * 8) = v;
+    }
   }
 }
 
 __global__ void k_nhwc2nchw(const bf16_t* __restrict__ x,
                             float* __restrict__ y, long NHW, long HW, int C,
                             int Cp) {
-  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
-       p += (long)gridDim.x * blockDim.x) {
-    const long n = p / HW, rem = p - n * HW;
-    const bf16_t* src = x + p * Cp;
-    float* dst = y + (n * C) * HW + rem;
-    for (int c = 0; c < C; ++c) dst[(long)c * HW] = bf2f(src[c]);
+  __shared__ float tile[16][257];
+  const int tid = threadIdx.x;
+  const long p = (long)blockIdx.x * 256 + tid;
+  const int cc0 = blockIdx.z * 16;
+  const bool pv = p < NHW;
+  if (pv) {
+#pragma unroll
+    for (int v8 = 0; v8 < 2; ++v8) {
+      const bf16x8 v =
+          *reinterpret_cast<const bf16x8*>(x + p * Cp + cc0 + v8 * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tile[v8 * 8 + e][tid] = bf2f(v[e]);
+    }
+  }
+  __syncthreads();
+  const long n = pv ? p / HW : 0;
+  const long rem = p - n * HW;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const int c = cc0 + i;
+    if (pv && c < C) y[(n * C + c) * HW + rem] = tile[i][tid];
   }
 }
 
@@ -352,7 +384,9 @@ at::Tensor nchw_to_nhwc(const at::Tensor& x, int64_t Cp) {
   const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   auto y = at::empty({N, H, W, Cp}, x.options().dtype(at::kBFloat16));
   const long NHW = N * H * W;
-  hipLaunchKernelGGL(k_nchw2nhwc, dim3(grid1d(NHW)), dim3(TPB), 0,
+  hipLaunchKernelGGL(k_nchw2nhwc,
+                     dim3((unsigned)((NHW + 255) / 256), 1, (unsigned)(Cp / 16)),
+                     dim3(256), 0,
                      cur_stream(), x.contiguous().data_ptr<float>(),
                      (bf16_t*)y.data_ptr(), NHW, H * W, (int)C, (int)Cp);
   HIP_CHECK_LAST();
@@ -364,7 +398,9 @@ at::Tensor nhwc_to_nchw(const at::Tensor& x, int64_t C) {
   const long N = x.size(0), H = x.size(1), W = x.size(2), Cp = x.size(3);
   auto y = at::empty({N, C, H, W}, x.options().dtype(at::kFloat));
   const long NHW = N * H * W;
-  hipLaunchKernelGGL(k_nhwc2nchw, dim3(grid1d(NHW)), dim3(TPB), 0,
+  hipLaunchKernelGGL(k_nhwc2nchw,
+                     dim3((unsigned)((NHW + 255) / 256), 1, (unsigned)(Cp / 16)),
+                     dim3(256), 0,
                      cur_stream(), (const bf16_t*)x.contiguous().data_ptr(),
                      y.data_ptr<float>(), NHW, H * W, (int)C, (int)Cp);
   HIP_CHECK_LAST();
