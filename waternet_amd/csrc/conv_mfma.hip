@@ -290,6 +290,22 @@ __global__ void k_splitk_finalize(const float* __restrict__ Y32,
 constexpr int TR_MBS = 96;    // mblk stride (elements)
 constexpr int TR_KBS = 1552;  // kblk stride (elements)
 
+// Magic-multiply unsigned division by a launch-constant divisor d:
+// q = (n * mul) >> 42 with mul = floor(2^42/d)+1 — exact for n*d < 2^42
+// (holds for any image geometry we launch). Replaces the 64-bit v_div
+// chains the m->(n,oy,ox) decode otherwise costs per chunk (PMC r06:
+// wgrad SQ_WAIT_INST_ANY 47-75% with zero bank conflicts = serial VALU
+// division RAW stalls).
+struct MagicDiv {
+  unsigned long long mul;
+  static MagicDiv make(unsigned d) {
+    return {(0x40000000000ULL / d) + 1};
+  }
+};
+WN_DEVFN unsigned magic_div(unsigned n, unsigned long long mul) {
+  return (unsigned)(((unsigned long long)n * mul) >> 42);
+}
+
 // elem offset of (m, col) inside a tr image
 WN_DEVFN int tr_addr(int m, int col) {
   return (col >> 4) * TR_KBS + (m >> 2) * TR_MBS + (m & 3) * 16 + (col & 15);
@@ -315,7 +331,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     const bf16_t* __restrict__ X,   // (N,H,W,Cp)
     float* __restrict__ dW,         // (K, C, KS, KS) fp32, pre-zeroed
     int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C,
-    int splitm) {
+    int splitm, unsigned long long mulHW, unsigned long long mulW) {
   static_assert(WN_MFMA_KMAP == 0, "wgrad staging assumes KMAP 0");
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;
@@ -393,13 +409,14 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
       bf16x8 v = {};
       long m = m0 + bMl[s];
       if (m < M && bIn[s]) {
-        int n = (int)(m / HW);
-        int rem = (int)(m - (long)n * HW);
-        int oy = rem / W, ox = rem - (rem / W) * W;
-        int iy = oy + bDy[s] - PAD, ix = ox + bDx[s] - PAD;
+        unsigned n = magic_div((unsigned)m, mulHW);
+        unsigned rem = (unsigned)m - n * (unsigned)HW;
+        unsigned oy = magic_div(rem, mulW);
+        int ox = (int)(rem - oy * (unsigned)W);
+        int iy = (int)oy + bDy[s] - PAD, ix = ox + bDx[s] - PAD;
         if (iy >= 0 && iy < H && ix >= 0 && ix < W)
           v = *reinterpret_cast<const bf16x8*>(
-              X + (((long)(n * H + iy) * W + ix) << log2Cp) + bC[s]);
+              X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + bC[s]);
       }
       bR[s] = v;
     }
@@ -790,7 +807,9 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
     hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV>), dim3(gx, gy, split),
                        dim3(256), lds, stream, (const bf16_t*)dy.data_ptr(),
                        (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
-                       H, W, Cp, log2i(Cp), Kp, K, C, split);
+                       H, W, Cp, log2i(Cp), Kp, K, C, split,
+                       MagicDiv::make((unsigned)(H * W)).mul,
+                       MagicDiv::make((unsigned)W).mul);
   };
   auto launch_ks = [&](auto ks_const) {
     if (BK == 128)
