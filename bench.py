@@ -34,7 +34,56 @@ def parse_args():
     p.add_argument("--width", type=int, default=112)
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph capture of the train step")
+    p.add_argument("--mode", choices=["train", "infer1080p"], default="train",
+                   help="train = flagship training step (driver metric); "
+                        "infer1080p = BASELINE config 4, hipGraph-captured "
+                        "bs=1 1080p per-frame video inference")
     return p.parse_args()
+
+
+def infer1080p(args):
+    """BASELINE config 4: 1080p (1920x1088, /32-padded as the reference
+    requires multiples of 32 — training_utils.py:99-103) bs=1 per-frame
+    pipeline, hipGraph-captured: H2D -> GPU preprocess -> forward -> fused
+    u8 postprocess -> D2H."""
+    import numpy as np
+
+    from waternet_amd.engine.inferencer import InferenceEngine
+    from waternet_amd.models.waternet import WaterNet
+
+    h, w = 1088, 1920
+    torch.manual_seed(0)
+    model = WaterNet().to("cuda:0")
+    eng = InferenceEngine(model, h, w, device="cuda:0",
+                          use_graph=not args.no_graph)
+    rng = np.random.default_rng(0)
+    frames = [rng.integers(0, 256, size=(h, w, 3), dtype=np.uint8)
+              for _ in range(4)]
+    for i in range(max(args.warmup, 2)):
+        eng.infer_frame(frames[i % len(frames)])
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        eng.infer_frame(frames[i % len(frames)])
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(json.dumps({
+        "metric": "video inference frames/sec at 1080p bs=1 (hipGraph)",
+        "value": args.steps / dt,
+        "unit": "frames/sec",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "config": {"model": "waternet", "global_batch": 1,
+                   "im_size": f"{h}x{w}", "parallelism": "single",
+                   "graph": not args.no_graph},
+    }))
 
 
 def main():
@@ -45,6 +94,10 @@ def main():
 
     if not torch.cuda.is_available():
         raise SystemExit("bench.py requires a ROCm GPU")
+
+    if args.mode == "infer1080p":
+        infer1080p(args)
+        return
 
     torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)

@@ -99,6 +99,33 @@ def test_inference_engine_matches_cpu_reference():
         diff.mean(), diff.max())
 
 
+def test_highres_train_step():
+    """BASELINE config 5 path: one 512x512 training step through the native
+    kernels (small batch here; bs=64/GPU is the 8-GPU stress config)."""
+    from waternet_amd.engine.fast import BenchTrainer
+
+    tr = BenchTrainer(batch_size=2, height=512, width=512, device=DEV,
+                      use_graph=False, seed=13, pool_size=1)
+    tr.step()
+    torch.cuda.synchronize()
+    m = tr.metrics()
+    assert all(np.isfinite(v) for v in m.values()), m
+
+
+def test_inference_engine_1080p_shape():
+    """BASELINE config 4 path: one 1080p frame through the hipGraph engine."""
+    from waternet_amd.engine.inferencer import InferenceEngine
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(17)
+    model = WaterNet().to(DEV)
+    eng = InferenceEngine(model, 1088, 1920, device=DEV, use_graph=True)
+    rng = np.random.default_rng(17)
+    frame = rng.integers(0, 256, size=(1088, 1920, 3), dtype=np.uint8)
+    out = eng.infer_frame(frame)
+    assert out.shape == (1088, 1920, 3) and out.dtype == np.uint8
+
+
 def test_preprocess_112():
     """Full-size preprocess at the flagship 112x112 shape."""
     from waternet_amd.ops.preprocess import gpu_transform_batch
