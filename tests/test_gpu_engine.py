@@ -138,3 +138,27 @@ def test_preprocess_112():
     torch.cuda.synchronize()
     for t in (wb, gc, he):
         assert t.shape == raw.shape and t.dtype == torch.uint8
+
+
+def test_stream_join_ordering():
+    """Race-assertion helpers: copy-stream produce -> compute consume."""
+    from waternet_amd.utils.streams import StreamJoin, ordered_copy
+
+    cs = torch.cuda.Stream()
+    src = torch.randn(1 << 20, device=DEV)
+    dst = torch.empty_like(src)
+    ev = ordered_copy(dst, src, cs)
+    torch.cuda.current_stream().wait_event(ev)
+    assert torch.equal(dst, src)
+
+    join = StreamJoin(cs)
+    with torch.cuda.stream(cs):
+        dst.add_(1.0)
+    join.mark()
+    join.wait()
+    torch.cuda.synchronize()
+    assert torch.allclose(dst, src + 1.0)
+
+    fresh = StreamJoin(cs)
+    with pytest.raises(AssertionError):
+        fresh.wait()  # wait() without mark() must fail loudly
