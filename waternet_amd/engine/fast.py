@@ -57,9 +57,11 @@ class BenchTrainer:
             if pin:
                 raw, ref = raw.pin_memory(), ref.pin_memory()
             self.pool.append((raw, ref))
-        # static input buffers (graph-capture safe)
-        self.raw_static = torch.empty_like(self.pool[0][0])
-        self.ref_static = torch.empty_like(self.pool[0][1])
+        # static DEVICE input buffers (graph-capture safe H2D targets)
+        self.raw_static = torch.empty(
+            (batch_size, height, width, 3), dtype=torch.uint8,
+            device=self.device)
+        self.ref_static = torch.empty_like(self.raw_static)
         # on-device metric accumulators
         self.metric_sums = torch.zeros(5, dtype=torch.float64,
                                        device=self.device)
