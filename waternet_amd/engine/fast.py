@@ -114,9 +114,11 @@ class BenchTrainer:
             # independent compute.
             with trace_range("allreduce"):
                 self.opt.grads.div_(self.world)
-                main = torch.cuda.current_stream()
-                self._comm_stream.wait_stream(main)
-                with torch.cuda.stream(self._comm_stream):
+                if self._comm_stream is not None:
+                    self._comm_stream.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(self._comm_stream):
+                        torch.distributed.all_reduce(self.opt.grads)
+                else:  # CPU/gloo path (tests)
                     torch.distributed.all_reduce(self.opt.grads)
 
         with trace_range("metrics"), torch.no_grad():
@@ -128,7 +130,7 @@ class BenchTrainer:
                 mse.detach().double(), ssim.double(), psnr.double()
             ])
 
-        if self.world > 1:
+        if self.world > 1 and self._comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self._comm_stream)
         with trace_range("optimizer"):
             self.opt.step()
