@@ -73,8 +73,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 x BM*64
-  bf16_t* lB = lA + 2 * BM * 64;                             // 2 x 2*LB
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 x BM*32
+  bf16_t* lB = lA + 2 * BM * 32;                             // 2 x LB
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -119,42 +119,35 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
 
   static_assert(WN_MFMA_KMAP == 0, "glds staging assumes KMAP 0");
 
-  // Stage one 64-deep K-tile (two 32-halves) straight into LDS buffer
-  // `buf` with global_load_lds (16 B per lane, lane-linear LDS image —
-  // guide §5 "common mistake 1": width-16 glds is the staging lever;
-  // 64-deep K per barrier is the guide's "fix BK first"). Out-of-range
-  // lanes (conv halo, channel pad, M tail, K tail) read from Zero16 — every
-  // lane always issues its DMA so the image is fully defined and the K
-  // tail needs no edge code (zero tiles accumulate nothing).
+  // Stage one K-step's A and B tiles straight into LDS buffer `buf` with
+  // global_load_lds (16 B per lane, lane-linear LDS image — guide §5
+  // "common mistake 1": width-16 glds is the staging lever). Out-of-range
+  // lanes (conv halo, channel pad, M tail) read from Zero16 instead —
+  // every lane always issues its DMA so the image is fully defined.
   auto stage = [&](int buf, int k0) {
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-#pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const bf16_t* src = Zero16;
-        int rsc = k0 + h * 32 + aKb[s] * 8;
-        if (aMv[s] && rsc < KG) {
-          int tap = rsc >> log2Cp;
-          int c = rsc & (Cp - 1);
-          int dy = tap / KS, dx = tap - (tap / KS) * KS;
-          int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
-          if (iy >= 0 && iy < H && ix >= 0 && ix < W)
-            src = X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c;
-        }
-        __builtin_amdgcn_global_load_lds(
-            src, lA + buf * (BM * 64) + h * (BM * 32) + (tid + s * 256) * 8,
-            16, 0, 0);
+    for (int s = 0; s < 2; ++s) {
+      const bf16_t* src = Zero16;
+      int rsc = k0 + aKb[s] * 8;
+      if (aMv[s] && rsc < KG) {
+        int tap = rsc >> log2Cp;
+        int c = rsc & (Cp - 1);
+        int dy = tap / KS, dx = tap - (tap / KS) * KS;
+        int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
+        if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+          src = X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c;
       }
+      __builtin_amdgcn_global_load_lds(
+          src, lA + buf * (BM * 32) + (tid + s * 256) * 8, 16, 0, 0);
+    }
 #pragma unroll
-      for (int s = 0; s < NBS; ++s) {
-        const bf16_t* src = Zero16;
-        int rsc = k0 + h * 32 + bKb[s] * 8;
-        if (bV[s] && bK[s] < Kp && rsc < KG)
-          src = Wp + (long)bK[s] * KG + rsc;
-        __builtin_amdgcn_global_load_lds(
-            src, lB + buf * (2 * LB) + h * LB + (tid + s * 256) * 8, 16, 0,
-            0);
-      }
+    for (int s = 0; s < NBS; ++s) {
+      const bf16_t* src = Zero16;
+      int rsc = k0 + bKb[s] * 8;
+      if (bV[s] && bK[s] < Kp && rsc < KG)
+        src = Wp + (long)bK[s] * KG + rsc;
+      __builtin_amdgcn_global_load_lds(
+          src, lB + buf * LB + (tid + s * 256) * 8, 16, 0, 0);
     }
   };
 
@@ -164,7 +157,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
 #pragma unroll
     for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int nkAll = (KG + 63) / 64;  // 64-deep K-tiles
+  const int nkAll = (KG + 31) / 32;
   // SPLITK: z-slice [ks0, nk) of the K loop; else the whole range. Floor
   // slicing keeps every slice non-empty for gz <= nkAll, so every fp32
   // slab is fully written (the finalize pass sums all of them).
@@ -173,41 +166,37 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     ks0 = (int)((long)blockIdx.z * nkAll / gridDim.z);
     nk = (int)((long)(blockIdx.z + 1) * nkAll / gridDim.z);
   }
-  stage(0, ks0 * 64);
+  stage(0, ks0 * 32);
   __syncthreads();  // drains the glds (vmcnt 0) + barrier
 
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
-  // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next K-tile's
-  // DMA first, ds_read + MFMA the current buffer (both 32-halves), one
-  // vmcnt(0)+barrier per 64-deep K-tile (__syncthreads drains the DMA).
+  // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next tile's
+  // DMA first, ds_read + MFMA the current buffer, one vmcnt(0)+barrier per
+  // K-step (the __syncthreads drains the in-flight DMA).
   int cur = 0;
   for (int ks = ks0; ks < nk; ++ks) {
-    if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 64);
+    if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
+    bf16x8 aF[FM], bF[FN];
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      bf16x8 aF[FM], bF[FN];
-#pragma unroll
-      for (int fm = 0; fm < FM; ++fm) {
-        int mfG = wr * FM + fm;
-        aF[fm] = *reinterpret_cast<const bf16x8*>(
-            lA + cur * (BM * 64) + h * (BM * 32) +
-            ((mfG * 4 + lg) * 16 + li) * 8);
-      }
-#pragma unroll
-      for (int fn = 0; fn < FN; ++fn) {
-        int nfG = wc * FN + fn;
-        bF[fn] = *reinterpret_cast<const bf16x8*>(
-            lB + cur * (2 * LB) + h * LB + ((nfG * 4 + lg) * 16 + li) * 8);
-      }
-#pragma unroll
-      for (int fm = 0; fm < FM; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < FN; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+    for (int fm = 0; fm < FM; ++fm) {
+      int mfG = wr * FM + fm;
+      aF[fm] = *reinterpret_cast<const bf16x8*>(
+          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
     }
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn) {
+      int nfG = wc * FN + fn;
+      bF[fn] = *reinterpret_cast<const bf16x8*>(
+          lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
+    }
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
     if (ks + 1 < nk) __syncthreads();
     cur ^= 1;
   }
@@ -712,7 +701,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
   const float* bptr =
       bias.has_value() ? bias->data_ptr<float>() : nullptr;
   const int gx = (int)((M + 127) / 128);
-  const int nk64 = (KS * KS * Cp + 63) / 64;  // 64-deep K-tiles
+  const int nk = (KS * KS * Cp + 31) / 32;
   // 16 B zero source for glds halo/pad lanes (per-device, created once)
   static thread_local at::Tensor zero16;
   if (!zero16.defined() || zero16.device() != x.device())
@@ -723,14 +712,13 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
     constexpr int NBS = (BN * 4 + 255) / 256;
-    const size_t lds = (2 * 128 * 64 + 4 * NBS * 256 * 8) * sizeof(bf16_t);
+    const size_t lds = (2 * 128 * 32 + 2 * NBS * 256 * 8) * sizeof(bf16_t);
     // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
     // 256 CUs idle; split the K loop across gz slices into fp32 partials,
     // then finalize bias+act+bf16 in a second tiny pass.
     const int base = gx * gy;
     int gz = 1;
-    if (base < 320 && nk64 >= 8)
-      gz = std::min(std::max(1, 512 / base), nk64 / 2);
+    if (base < 320 && nk >= 16) gz = std::min(std::max(1, 512 / base), nk / 4);
     if (gz > 1) {
       auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
       hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
