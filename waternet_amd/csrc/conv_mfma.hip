@@ -73,8 +73,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 3 x BM*32
-  bf16_t* lB = lA + 3 * BM * 32;                             // 3 x LB
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 x BM*32
+  bf16_t* lB = lA + 2 * BM * 32;                             // 2 x LB
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -166,47 +166,30 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     ks0 = (int)((long)blockIdx.z * nkAll / gridDim.z);
     nk = (int)((long)(blockIdx.z + 1) * nkAll / gridDim.z);
   }
+  stage(0, ks0 * 32);
+  __syncthreads();  // drains the glds (vmcnt 0) + barrier
+
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
-  // 3-buffer glds pipeline with COUNTED vmcnt + raw barriers (guide §5
-  // "Pipelining across barriers": leaving one tile's DMA in flight across
-  // each barrier instead of draining vmcnt(0) per step). Each wave issues
-  // GPS glds per stage; s_waitcnt vmcnt(GPS) at the step boundary waits
-  // for tile ks+1 while tile ks+2's DMA stays outstanding. Raw s_barrier
-  // only — __syncthreads would emit vmcnt(0) and drain the pipeline.
-  constexpr int GPS = 2 + NBS;  // glds per stage per thread (3 or 4)
-  auto wait_tiles_in_flight = [&](int inflight) {
-    if (inflight > 0) {
-      if constexpr (GPS == 3)
-        asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
-      else
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-  };
-
-  stage(0, ks0 * 32);
-  if (ks0 + 1 < nk) stage(1, (ks0 + 1) * 32);
-  wait_tiles_in_flight(ks0 + 1 < nk ? 1 : 0);  // tile ks0 landed
-  __builtin_amdgcn_s_barrier();
-
+  // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next tile's
+  // DMA first, ds_read + MFMA the current buffer, one vmcnt(0)+barrier per
+  // K-step (the __syncthreads drains the in-flight DMA).
+  int cur = 0;
   for (int ks = ks0; ks < nk; ++ks) {
-    const int b = (ks - ks0) % 3;
-    if (ks + 2 < nk) stage((b + 2) % 3, (ks + 2) * 32);
+    if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
     bf16x8 aF[FM], bF[FN];
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm) {
       int mfG = wr * FM + fm;
       aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + b * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
+          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
     }
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn) {
       int nfG = wc * FN + fn;
       bF[fn] = *reinterpret_cast<const bf16x8*>(
-          lB + b * LB + ((nfG * 4 + lg) * 16 + li) * 8);
+          lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
     }
 #pragma unroll
     for (int fm = 0; fm < FM; ++fm)
@@ -214,10 +197,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
       for (int fn = 0; fn < FN; ++fn)
         acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    if (ks + 1 < nk) {
-      wait_tiles_in_flight(ks + 2 < nk ? 1 : 0);  // tile ks+1 landed
-      __builtin_amdgcn_s_barrier();
-    }
+    if (ks + 1 < nk) __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: bias + activation (+ pad-channel zeroing), or fp32
@@ -731,7 +712,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
     constexpr int NBS = (BN * 4 + 255) / 256;
-    const size_t lds = (3 * 128 * 32 + 3 * NBS * 256 * 8) * sizeof(bf16_t);
+    const size_t lds = (2 * 128 * 32 + 2 * NBS * 256 * 8) * sizeof(bf16_t);
     // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
     // 256 CUs idle; split the K loop across gz slices into fp32 partials,
     // then finalize bias+act+bf16 in a second tiny pass.
