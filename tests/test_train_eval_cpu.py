@@ -128,3 +128,31 @@ def test_inference_cli_image(tmp_path):
     assert result.exists()
     with Image.open(result) as im:
         assert im.size == (64, 64)
+
+
+def test_train_resume_full_state(tmp_path, monkeypatch):
+    """--full-state writes a sidecar; --resume restores model + optimizer +
+    scheduler + epoch (the reference restarts the LR schedule on resume)."""
+    import json
+    import train as train_cli
+
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setattr(train_cli, "__file__", str(tmp_path / "train.py"))
+    train_cli.main(["--epochs", "2", "--batch-size", "2", "--height", "32",
+                    "--width", "32", "--synthetic", "6", "--full-state"])
+    savedir = tmp_path / "training" / "0"
+    assert (savedir / "last.pt").exists()
+    assert (savedir / "last-trainstate.pt").exists()
+    state = __import__("torch").load(savedir / "last-trainstate.pt")
+    assert state["epoch"] == 1
+
+    train_cli.main(["--epochs", "3", "--batch-size", "2", "--height", "32",
+                    "--width", "32", "--synthetic", "6", "--full-state",
+                    "--resume", str(savedir)])
+    savedir2 = tmp_path / "training" / "1"
+    state2 = __import__("torch").load(savedir2 / "last-trainstate.pt")
+    assert state2["epoch"] == 2  # only epoch index 2 ran
+    # metrics CSV holds exactly the resumed epochs
+    lines = (savedir2 / "metrics-train.csv").read_text().strip().splitlines()
+    assert len(lines) == 2  # header + 1 epoch
+    json.loads((savedir2 / "config.json").read_text())

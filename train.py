@@ -57,6 +57,10 @@ def parse_args(argv=None):
     parser.add_argument("--num-workers", type=int, default=0)
     parser.add_argument("--shuffle", action="store_true")
     parser.add_argument("--full-state", action="store_true")
+    parser.add_argument("--resume", type=str, default=None, metavar="DIR",
+                        help="resume from a savedir holding last.pt (+ "
+                             "last-trainstate.pt for optimizer/scheduler/"
+                             "epoch state written by --full-state)")
     return parser.parse_args(argv)
 
 
@@ -115,7 +119,10 @@ def main(argv=None):
         print(f"Using device: {device}")
 
     model = WaterNet()
-    if args.weights is not None:
+    if args.resume is not None:
+        model.load_state_dict(torch.load(Path(args.resume) / "last.pt",
+                                         map_location="cpu"))
+    elif args.weights is not None:
         with open(args.weights, "rb") as f:
             model.load_state_dict(torch.load(f, map_location="cpu"))
     model.to(device)
@@ -125,6 +132,19 @@ def main(argv=None):
     scheduler = torch.optim.lr_scheduler.StepLR(
         optimizer, step_size=10000, gamma=0.1
     )
+    start_epoch = 0
+    if args.resume is not None:
+        # full resume: optimizer/scheduler/epoch from the sidecar if present
+        # (the reference's resume via --weights restarts the LR schedule —
+        # SURVEY §5.3; the sidecar fixes that without changing last.pt)
+        sidecar = Path(args.resume) / "last-trainstate.pt"
+        if sidecar.exists():
+            state = torch.load(sidecar, map_location="cpu")
+            optimizer.load_state_dict(state["optimizer"])
+            scheduler.load_state_dict(state["scheduler"])
+            start_epoch = state["epoch"] + 1
+            if rank0:
+                print(f"Resumed epoch {start_epoch} from {sidecar}")
 
     vgg_model = PerceptualModel().to(device).eval()
 
@@ -138,7 +158,7 @@ def main(argv=None):
     saved_train = {k: [] for k in TRAIN_METRICS_NAMES}
     saved_val = {k: [] for k in VAL_METRICS_NAMES}
 
-    for epoch in range(args.epochs):
+    for epoch in range(start_epoch, args.epochs):
         train_metrics = train_one_epoch(
             model, train_loader, optimizer, scheduler, vgg_model, device,
             epoch_num=epoch, total_epochs=args.epochs,
