@@ -307,8 +307,9 @@ WN_DEVFN unsigned magic_div(unsigned n, unsigned long long mul) {
 }
 
 // elem offset of (m, col) inside a tr image
+template <int KBS = TR_KBS>
 WN_DEVFN int tr_addr(int m, int col) {
-  return (col >> 4) * TR_KBS + (m >> 2) * TR_MBS + (m & 3) * 16 + (col & 15);
+  return (col >> 4) * KBS + (m >> 2) * TR_MBS + (m & 3) * 16 + (col & 15);
 }
 
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
@@ -325,7 +326,7 @@ WN_DEVFN bf16x4 ds_tr16(unsigned addr) {
   return v;
 }
 
-template <int KS, int BK>
+template <int KS, int BK, int CH = 64>  // CH: m rows per chunk (32 or 64)
 __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     const bf16_t* __restrict__ dY,  // (N,H,W,Kp)
     const bf16_t* __restrict__ X,   // (N,H,W,Cp)
@@ -336,7 +337,9 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;
   constexpr int BR = 128;           // rsc tile
-  constexpr int CH = 64;            // m chunk
+  // per-chunk tr image kblk stride: CH/4 mblks x 96 elems + 16 pad, which
+  // keeps the stride 8 mod 32 dwords (conflict-free staging writes)
+  constexpr int KBS = CH / 4 * TR_MBS + 16;
   constexpr int WR = (BK >= 64) ? 2 : 1;   // wave rows (k dim)
   constexpr int WC = 4 / WR;               // wave cols (rsc dim)
   constexpr int FK = BK / WR / 16;  // k fragments per wave
@@ -349,8 +352,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const int HW = H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // (BK/16) x TR_KBS
-  bf16_t* lB = lA + (BK / 16) * TR_KBS;           // (BR/16) x TR_KBS
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // (BK/16) x KBS
+  bf16_t* lB = lA + (BK / 16) * KBS;              // (BR/16) x KBS
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -425,13 +428,14 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 #pragma unroll
     for (int s = 0; s < ASLOT; ++s) {
       if (!aV[s]) continue;
-      *reinterpret_cast<bf16x8*>(lA + tr_addr(aMl[s], aJg[s] * 8)) = aR[s];
+      *reinterpret_cast<bf16x8*>(
+          lA + tr_addr<KBS>(aMl[s], aJg[s] * 8)) = aR[s];
     }
 #pragma unroll
     for (int s = 0; s < BSLOT; ++s) {
       int slot = tid + s * 256;
       int jg = slot % (BR / 8);
-      *reinterpret_cast<bf16x8*>(lB + tr_addr(bMl[s], jg * 8)) = bR[s];
+      *reinterpret_cast<bf16x8*>(lB + tr_addr<KBS>(bMl[s], jg * 8)) = bR[s];
     }
   };
 
@@ -469,14 +473,14 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     for (int s2 = 0; s2 < CH / 32; ++s2) {
 #pragma unroll
       for (int f = 0; f < FK; ++f) {
-        const unsigned base = aTr0 + (unsigned)((wr * FK + f) * TR_KBS +
+        const unsigned base = aTr0 + (unsigned)((wr * FK + f) * KBS +
                                                 s2 * 8 * TR_MBS) * 2;
         aT[s2][f][0] = ds_tr16(base);
         aT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
       }
 #pragma unroll
       for (int f = 0; f < FR; ++f) {
-        const unsigned base = bTr0 + (unsigned)((wc * FR + f) * TR_KBS +
+        const unsigned base = bTr0 + (unsigned)((wc * FR + f) * KBS +
                                                 s2 * 8 * TR_MBS) * 2;
         bT[s2][f][0] = ds_tr16(base);
         bT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
@@ -794,17 +798,20 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   TORCH_CHECK(dw.size(2) == ks && dw.size(3) == ks);
   const int KG = (int)(ks * ks) * Cp;
   const int BK = std::min(Kp, 128);
+  constexpr int WG_CH = 64;  // m rows per chunk (32 measured within noise
+                             // of 64 at bs=16 112^2 despite 2x occupancy)
   const int gx = (Kp + BK - 1) / BK, gy = (KG + 127) / 128;
-  // split so gx*gy*split fills 256 CUs x 2 resident blocks
+  // split so gx*gy*split fills 256 CUs x ~3 resident blocks
   int split = std::max(1, 640 / std::max(1, gx * gy));
-  const long nChunks = ((long)N * H * W + 63) / 64;
+  const long nChunks = ((long)N * H * W + WG_CH - 1) / WG_CH;
   split = (int)std::min<long>(split, nChunks);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
-  const size_t lds = (size_t)(BK / 16 + 128 / 16) * TR_KBS * sizeof(bf16_t);
+  const size_t lds =
+      (size_t)(BK / 16 + 128 / 16) * (WG_CH / 4 * 96 + 16) * sizeof(bf16_t);
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
-    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV>), dim3(gx, gy, split),
+    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, WG_CH>), dim3(gx, gy, split),
                        dim3(256), lds, stream, (const bf16_t*)dy.data_ptr(),
                        (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
                        H, W, Cp, log2i(Cp), Kp, K, C, split,
