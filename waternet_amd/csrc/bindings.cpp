@@ -12,6 +12,7 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
 void bias_grad(const at::Tensor& dy, at::Tensor& db);
 at::Tensor pack_weight_fwd(const at::Tensor& w, int64_t Kp, int64_t Cp);
 at::Tensor pack_weight_dgrad(const at::Tensor& w, int64_t Kp, int64_t Cp);
+void pack_all(const at::Tensor& desc, int64_t njobs);
 at::Tensor probe_tr();
 at::Tensor probe_tr_raw();
 
@@ -61,6 +62,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_grad", &bias_grad, "bias grad (column sum)");
   m.def("pack_weight_fwd", &pack_weight_fwd);
   m.def("pack_weight_dgrad", &pack_weight_dgrad);
+  m.def("pack_all", &pack_all, "batched fwd+dgrad weight repack");
   m.def("probe_tr", &probe_tr, "ds_read_b64_tr_b16 lane-mapping probe");
   m.def("probe_tr_raw", &probe_tr_raw, "raw tr permutation dump");
   m.def("build_inputs", &build_inputs);

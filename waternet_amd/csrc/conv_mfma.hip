@@ -604,10 +604,12 @@ at::Tensor probe_tr_raw() {
 
 // Each thread owns one 8-wide k-group and vector-loads bf16x8 rows of dY,
 // accumulating 8 fp32 partials in registers; threads covering the same
-// k-group at different m-phases reduce through LDS, then one atomicAdd per
-// (block, k). Grid: (ceil(Kp/8/KGR), msplit) with KGR k-groups per block.
+// k-group at different m-phases reduce through LDS, then each block writes
+// a plain fp32 partial slab Part[blockIdx.y][Kp] (no atomics — dB is a few
+// cache lines and msplit*K atomics serialize on them; measured 2.2 ms ->
+// slabs + the tiny reduce below). Grid: (ceil(ngrp/KGR), msplit).
 __global__ void k_bias_grad(const bf16_t* __restrict__ dY,
-                            float* __restrict__ dB, long M, int Kp, int K,
+                            float* __restrict__ Part, long M, int Kp,
                             int msplit) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* red = reinterpret_cast<float*>(smem);  // [256][8] fp32
@@ -638,10 +640,23 @@ __global__ void k_bias_grad(const bf16_t* __restrict__ dY,
     for (int r = 1; r < MR; ++r)
 #pragma unroll
       for (int e = 0; e < 8; ++e) acc[e] += red[(r * KGR + kg) * 8 + e];
-#pragma unroll
-    for (int e = 0; e < 8; ++e)
-      if (k0 + e < K) atomicAdd(&dB[k0 + e], acc[e]);
+    *reinterpret_cast<f32x4*>(&Part[(long)blockIdx.y * Kp + k0]) =
+        f32x4{acc[0], acc[1], acc[2], acc[3]};
+    *reinterpret_cast<f32x4*>(&Part[(long)blockIdx.y * Kp + k0 + 4]) =
+        f32x4{acc[4], acc[5], acc[6], acc[7]};
   }
+}
+
+// Reduce the [msplit][Kp] partials into dB (accumulating, like the old
+// atomic path: dB may hold a pre-existing gradient).
+__global__ void k_bias_grad_reduce(const float* __restrict__ Part,
+                                   float* __restrict__ dB, int Kp, int K,
+                                   int msplit) {
+  const int k = blockIdx.x * blockDim.x + threadIdx.x;
+  if (k >= K) return;
+  float s = 0.f;
+  for (int r = 0; r < msplit; ++r) s += Part[(long)r * Kp + k];
+  dB[k] += s;
 }
 
 // ---------------------------------------------------------------------------
@@ -680,6 +695,55 @@ __global__ void k_pack_dgrad(const float* __restrict__ Wm,  // (K,C,R,S)
     float v = (c < C && k < K) ? Wm[(((long)k * C + c) * R + r) * S + s] : 0.f;
     out[idx] = f2bf(v);
   }
+}
+
+// Batched repack: one launch packs EVERY layer's fwd + dgrad layouts from
+// the fp32 masters (the per-layer k_pack_* launches cost ~5 us each x 36
+// per training step under the lazy per-ConvSpec refresh). desc: int64
+// [njobs][9] = (Wm, wp, wd, K, C, R, S, Kp, Cp).
+__global__ void k_pack_all(const long* __restrict__ desc, int njobs) {
+  const long tid0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long nthreads = (long)gridDim.x * blockDim.x;
+  for (int j = 0; j < njobs; ++j) {
+    const long* d = desc + (long)j * 9;
+    const float* Wm = reinterpret_cast<const float*>(d[0]);
+    bf16_t* wp = reinterpret_cast<bf16_t*>(d[1]);
+    bf16_t* wd = reinterpret_cast<bf16_t*>(d[2]);
+    const int K = (int)d[3], C = (int)d[4], R = (int)d[5], S = (int)d[6];
+    const int Kp = (int)d[7], Cp = (int)d[8];
+    const int KGf = R * S * Cp;
+    const long tf = (long)Kp * KGf;
+    for (long idx = tid0; idx < tf; idx += nthreads) {
+      int k = (int)(idx / KGf);
+      int rsc = (int)(idx - (long)k * KGf);
+      int tap = rsc / Cp, c = rsc - (rsc / Cp) * Cp;
+      int r = tap / S, ss = tap - (tap / S) * S;
+      float v =
+          (k < K && c < C) ? Wm[(((long)k * C + c) * R + r) * S + ss] : 0.f;
+      wp[idx] = f2bf(v);
+    }
+    const int KGd = R * S * Kp;
+    const long td = (long)Cp * KGd;
+    for (long idx = tid0; idx < td; idx += nthreads) {
+      int c = (int)(idx / KGd);
+      int rsk = (int)(idx - (long)c * KGd);
+      int tap = rsk / Kp, k = rsk - (rsk / Kp) * Kp;
+      int rr = tap / S, sss = tap - (tap / S) * S;
+      int r = R - 1 - rr, s2 = S - 1 - sss;  // rotate 180
+      float v =
+          (c < C && k < K) ? Wm[(((long)k * C + c) * R + r) * S + s2] : 0.f;
+      wd[idx] = f2bf(v);
+    }
+  }
+}
+
+void pack_all(const at::Tensor& desc, int64_t njobs) {
+  TORCH_CHECK(desc.is_cuda() && desc.dtype() == at::kLong &&
+              desc.is_contiguous());
+  hipStream_t stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(k_pack_all, dim3(512), dim3(256), 0, stream,
+                     desc.data_ptr<long>(), (int)njobs);
+  HIP_CHECK_LAST();
 }
 
 // ---------------------------------------------------------------------------
@@ -723,6 +787,7 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     const int base = gx * gy;
     int gz = 1;
     if (base < 320 && nk >= 16) gz = std::min(std::max(1, 512 / base), nk / 4);
+    gz = std::min(gz, 8);  // finalize reads gz slabs; >8 is latency-negative
     if (gz > 1) {
       auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
       hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
@@ -847,16 +912,19 @@ void bias_grad(const at::Tensor& dy, at::Tensor& db) {
   const int ngrp = Kp / 8;
   const int KGR = std::min(ngrp, 32);
   const int gx = (ngrp + KGR - 1) / KGR;
-  // Modest split: dB has only K<=512 fp32 entries (a few cache lines), so
-  // msplit*K atomics serialize on them — 64 blocks saturate HBM read bw
-  // for these sizes while keeping atomic pressure ~64 per address.
-  int msplit = std::max(1, 64 / gx);
+  int msplit = std::max(1, 448 / gx);  // fill the chip; partial slabs
   const int MR = 256 / KGR;
   msplit = (int)std::min<long>(msplit, (M + MR - 1) / MR);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
+  auto part = at::empty({msplit, (long)Kp},
+                        dy.options().dtype(at::kFloat));
   hipLaunchKernelGGL(k_bias_grad, dim3(gx, msplit), dim3(256),
                      256 * 8 * sizeof(float), stream,
-                     (const bf16_t*)dy.data_ptr(), db.data_ptr<float>(), M,
+                     (const bf16_t*)dy.data_ptr(), part.data_ptr<float>(),
+                     M, Kp, msplit);
+  HIP_CHECK_LAST();
+  hipLaunchKernelGGL(k_bias_grad_reduce, dim3((K + 255) / 256), dim3(256),
+                     0, stream, part.data_ptr<float>(), db.data_ptr<float>(),
                      Kp, K, msplit);
   HIP_CHECK_LAST();
 }

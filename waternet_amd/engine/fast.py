@@ -72,11 +72,36 @@ class BenchTrainer:
         self._last_lr = lr
         self._comm_stream = (torch.cuda.Stream()
                              if self.device.type == "cuda" else None)
+        self._pack_desc = None
+        self._pack_specs = None
+
+    def _repack_weights(self):
+        """Batched fwd+dgrad repack of every WaterNet layer in ONE kernel
+        (vs ~36 lazy per-layer pack launches per step). No-op until the
+        native ConvSpec table exists (first forward builds it)."""
+        if self._pack_desc is None:
+            st = getattr(self.model, "_wn_native_state", None)
+            if st is None:
+                return
+            from waternet_amd.ops.conv import build_pack_descriptor
+
+            specs = list(st.cmg_specs)
+            for sl in st.refiner_specs.values():
+                specs.extend(sl)
+            self._pack_specs = specs
+            self._pack_desc = build_pack_descriptor(specs, self.device)
+        from waternet_amd.ops import ext
+        from waternet_amd.ops.conv import mark_specs_packed
+
+        ext().pack_all(self._pack_desc, len(self._pack_specs))
+        mark_specs_packed(self._pack_specs)
 
     # ---- one full training step on the current stream ----
     def _run_step_body(self):
         from waternet_amd.utils.profiling import trace_range
 
+        if self.device.type == "cuda":
+            self._repack_weights()
         raw_u8, ref_u8 = self.raw_static, self.ref_static
         with trace_range("preprocess"):
             wb_u8, gc_u8, he_u8 = gpu_transform_batch(raw_u8)

@@ -112,3 +112,28 @@ class ConvBiasAct(torch.autograd.Function):
 
 def conv_bias_act(x, spec: ConvSpec):
     return ConvBiasAct.apply(x, spec.mod.weight, spec.mod.bias, spec)
+
+
+def build_pack_descriptor(specs, device):
+    """int64 [n][9] descriptor for ext().pack_all over a list of ConvSpecs.
+    Forces allocation of each spec's packed buffers first (pointers must be
+    stable — they are: buffers are allocated once and rewritten in place)."""
+    import torch as _torch
+
+    rows = []
+    for s in specs:
+        s.refresh_if_needed()
+        w = s.mod.weight
+        rows.append([
+            w.data.data_ptr(), s._wp.data_ptr(), s._wd.data_ptr(),
+            s.K, s.C, s.ks, s.ks, s.Kp, s.Cp,
+        ])
+    return _torch.tensor(rows, dtype=_torch.int64, device=device)
+
+
+def mark_specs_packed(specs):
+    """Record current weight versions on every spec so the lazy per-layer
+    refresh skips (the batched pack_all just rewrote the buffers)."""
+    for s in specs:
+        w = s.mod.weight
+        s._version = (w._version, w.data_ptr())
