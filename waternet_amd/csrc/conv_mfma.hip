@@ -648,15 +648,23 @@ __global__ void k_bias_grad(const bf16_t* __restrict__ dY,
 }
 
 // Reduce the [msplit][Kp] partials into dB (accumulating, like the old
-// atomic path: dB may hold a pre-existing gradient).
+// atomic path: dB may hold a pre-existing gradient). One block per output
+// column, 256 threads strided over the msplit rows + LDS tree reduce.
 __global__ void k_bias_grad_reduce(const float* __restrict__ Part,
                                    float* __restrict__ dB, int Kp, int K,
                                    int msplit) {
-  const int k = blockIdx.x * blockDim.x + threadIdx.x;
-  if (k >= K) return;
+  __shared__ float red[256];
+  const int k = blockIdx.x;
   float s = 0.f;
-  for (int r = 0; r < msplit; ++r) s += Part[(long)r * Kp + k];
-  dB[k] += s;
+  for (int r = threadIdx.x; r < msplit; r += 256)
+    s += Part[(long)r * Kp + k];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) dB[k] += red[0];
 }
 
 // ---------------------------------------------------------------------------
@@ -923,9 +931,9 @@ void bias_grad(const at::Tensor& dy, at::Tensor& db) {
                      (const bf16_t*)dy.data_ptr(), part.data_ptr<float>(),
                      M, Kp, msplit);
   HIP_CHECK_LAST();
-  hipLaunchKernelGGL(k_bias_grad_reduce, dim3((K + 255) / 256), dim3(256),
-                     0, stream, part.data_ptr<float>(), db.data_ptr<float>(),
-                     Kp, K, msplit);
+  hipLaunchKernelGGL(k_bias_grad_reduce, dim3(K), dim3(256), 0, stream,
+                     part.data_ptr<float>(), db.data_ptr<float>(), Kp, K,
+                     msplit);
   HIP_CHECK_LAST();
 }
 
