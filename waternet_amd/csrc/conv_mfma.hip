@@ -287,8 +287,11 @@ __global__ void k_splitk_finalize(const float* __restrict__ Y32,
 // Requires the KMAP-0 fragment layout (frag_k16(g,e) = g*8+e).
 // ---------------------------------------------------------------------------
 
-constexpr int TR_MBS = 96;    // mblk stride (elements)
-constexpr int TR_KBS = 1552;  // kblk stride (elements)
+constexpr int TR_MBS = 72;    // mblk stride (36 dwords; 2 mblks = 8 mod 64
+                              // banks -> at most 2-way tr-read overlap, and
+                              // the image fits 4 blocks/CU at CH=64)
+constexpr int TR_KBS = 16 * TR_MBS + 16;  // kblk stride (584 dw = 8 mod 32:
+                                          // conflict-free staging writes)
 
 // Magic-multiply unsigned division by a launch-constant divisor d:
 // q = (n * mul) >> 42 with mul = floor(2^42/d)+1 — exact for n*d < 2^42
@@ -339,7 +342,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   constexpr int BR = 128;           // rsc tile
   // per-chunk tr image kblk stride: CH/4 mblks x 96 elems + 16 pad, which
   // keeps the stride 8 mod 32 dwords (conflict-free staging writes)
-  constexpr int KBS = CH / 4 * TR_MBS + 16;
+  constexpr int KBS = CH / 4 * TR_MBS + 16;  // = TR_KBS at CH=64
   constexpr int WR = (BK >= 64) ? 2 : 1;   // wave rows (k dim)
   constexpr int WC = 4 / WR;               // wave cols (rsc dim)
   constexpr int FK = BK / WR / 16;  // k fragments per wave
@@ -880,7 +883,7 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   split = (int)std::min<long>(split, nChunks);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
   const size_t lds =
-      (size_t)(BK / 16 + 128 / 16) * (WG_CH / 4 * 96 + 16) * sizeof(bf16_t);
+      (size_t)(BK / 16 + 128 / 16) * (WG_CH / 4 * 72 + 16) * sizeof(bf16_t);
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
