@@ -166,39 +166,85 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     ks0 = (int)((long)blockIdx.z * nkAll / gridDim.z);
     nk = (int)((long)(blockIdx.z + 1) * nkAll / gridDim.z);
   }
-  stage(0, ks0 * 32);
-  __syncthreads();  // drains the glds (vmcnt 0) + barrier
-
   const int lg = lane >> 4;   // fragment k-group
   const int li = lane & 15;   // fragment row/col
 
-  // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next tile's
-  // DMA first, ds_read + MFMA the current buffer, one vmcnt(0)+barrier per
-  // K-step (the __syncthreads drains the in-flight DMA).
-  int cur = 0;
-  for (int ks = ks0; ks < nk; ++ks) {
-    if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
-    bf16x8 aF[FM], bF[FN];
+  if constexpr (SPLITK) {
+    // Split-K shapes run at ~1 block/CU (small M, chip filled by K
+    // slices): the per-step vmcnt(0) drain is fully exposed there, so use
+    // a 3-buffer glds pipeline with COUNTED vmcnt + raw barriers (guide
+    // §5 "Pipelining across barriers": +40-83% in the 1-block/CU regime;
+    // measured NEGATIVE at the big-M shapes' 5-blocks/CU occupancy, which
+    // keep the simpler 2-phase below).
+    constexpr int GPS = 2 + NBS;  // glds per stage per thread (3 or 4)
+    auto wait_tiles = [&](int inflight) {
+      if (inflight > 0) {
+        if constexpr (GPS == 3)
+          asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+    };
+    stage(0, ks0 * 32);
+    if (ks0 + 1 < nk) stage(1, (ks0 + 1) * 32);
+    wait_tiles(ks0 + 1 < nk ? 1 : 0);  // tile ks0 landed
+    __builtin_amdgcn_s_barrier();
+    for (int ks = ks0; ks < nk; ++ks) {
+      const int b = (ks - ks0) % 3;
+      if (ks + 2 < nk) stage((b + 2) % 3, (ks + 2) * 32);
+      bf16x8 aF[FM], bF[FN];
 #pragma unroll
-    for (int fm = 0; fm < FM; ++fm) {
-      int mfG = wr * FM + fm;
-      aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
-    }
-#pragma unroll
-    for (int fn = 0; fn < FN; ++fn) {
-      int nfG = wc * FN + fn;
-      bF[fn] = *reinterpret_cast<const bf16x8*>(
-          lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
-    }
-#pragma unroll
-    for (int fm = 0; fm < FM; ++fm)
+      for (int fm = 0; fm < FM; ++fm)
+        aF[fm] = *reinterpret_cast<const bf16x8*>(
+            lA + b * (BM * 32) + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
 #pragma unroll
       for (int fn = 0; fn < FN; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    if (ks + 1 < nk) __syncthreads();
-    cur ^= 1;
+        bF[fn] = *reinterpret_cast<const bf16x8*>(
+            lB + b * LB + (((wc * FN + fn) * 4 + lg) * 16 + li) * 8);
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+      if (ks + 1 < nk) {
+        wait_tiles(ks + 2 < nk ? 1 : 0);  // tile ks+1 landed
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+  } else {
+    // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next
+    // tile's DMA first, ds_read + MFMA the current buffer, one
+    // vmcnt(0)+barrier per K-step (the __syncthreads drains the DMA).
+    stage(0, ks0 * 32);
+    __syncthreads();
+    int cur = 0;
+    for (int ks = ks0; ks < nk; ++ks) {
+      if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
+      bf16x8 aF[FM], bF[FN];
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm) {
+        int mfG = wr * FM + fm;
+        aF[fm] = *reinterpret_cast<const bf16x8*>(
+            lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
+      }
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn) {
+        int nfG = wc * FN + fn;
+        bF[fn] = *reinterpret_cast<const bf16x8*>(
+            lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
+      }
+#pragma unroll
+      for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+      if (ks + 1 < nk) __syncthreads();
+      cur ^= 1;
+    }
   }
 
   // ---- epilogue: bias + activation (+ pad-channel zeroing), or fp32
@@ -801,8 +847,10 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     gz = std::min(gz, 8);  // finalize reads gz slabs; >8 is latency-negative
     if (gz > 1) {
       auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
+      const size_t lds3 =
+          (3 * 128 * 32 + 3 * NBS * 256 * 8) * sizeof(bf16_t);
       hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
-                         dim3(256), lds, stream,
+                         dim3(256), lds3, stream,
                          (const bf16_t*)x.data_ptr(),
                          (const bf16_t*)wp.data_ptr(), bptr,
                          (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
