@@ -73,8 +73,10 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // 2 x BM*32
-  bf16_t* lB = lA + 2 * BM * 32;                             // 2 x LB
+  // SPLITK uses a 3-deep buffer ring (see below), the 2-phase path 2.
+  constexpr int NBUF = SPLITK ? 3 : 2;
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // NBUF x BM*32
+  bf16_t* lB = lA + NBUF * BM * 32;                          // NBUF x LB
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
