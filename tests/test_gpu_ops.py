@@ -60,6 +60,28 @@ def test_conv_fwd_parity(ext, ks, C, K):
     assert y[..., K:].abs().max().item() == 0.0 if Kp > K else True
 
 
+@pytest.mark.parametrize("ks,C,K", [(5, 128, 128), (7, 64, 64),
+                                    (3, 64, 128)])
+def test_conv_fwd_parity_bigM(ext, ks, C, K):
+    """Big-M shapes dispatch to the 8-wave 256-row phase-split kernel
+    (M >= 16384); verify parity there too."""
+    from waternet_amd.ops.conv import pow2_channels
+
+    torch.manual_seed(4)
+    N, H, W = 2, 96, 96  # M = 18432 >= 16384
+    x = torch.rand(N, C, H, W, device=DEV)
+    w = torch.randn(K, C, ks, ks, device=DEV) * 0.05
+    b = torch.randn(K, device=DEV) * 0.1
+    Cp, Kp = pow2_channels(C), pow2_channels(K)
+    wp = ext.pack_weight_fwd(w.contiguous(), Kp, Cp)
+    y = ext.conv2d_fwd(to_nhwc_bf16(x, Cp), wp, b, ks, Kp, K, 1)
+    got = from_nhwc(y, K)
+    ref = F.relu(F.conv2d(q(x), q(w), b, padding=ks // 2))
+    err = (got - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 2e-2, f"rel err {err/scale}"
+
+
 def test_conv_fwd_sigmoid(ext):
     from waternet_amd.ops.conv import pow2_channels
 

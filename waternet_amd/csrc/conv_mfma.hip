@@ -280,6 +280,203 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 8-wave 256xBN phase-split conv igemm for BIG-M shapes (guide §5 T3+T5:
+// wave role diversity + setprio + counted vmcnt; the 4-wave 2-phase kernel
+// above is schedule-bound at ~480 TF on these shapes — PMC r06/r18: ~58%
+// WAIT_ANY with MFMA, TA and HBM all unsaturated).
+//   - 512 threads = 8 waves as 4(M)x2(N); per-wave output 64 x BN/2.
+//   - 3-deep LDS ring, glds staging (GPS=3 per thread per tile), counted
+//     s_waitcnt vmcnt leaves the next tile's DMA in flight across barriers.
+//   - two MFMA phases per K-step with s_setprio(1) around the matrix math.
+// ---------------------------------------------------------------------------
+
+template <int BN, int KS>
+__global__ __launch_bounds__(512, 1) void k_conv_igemm8(
+    const bf16_t* __restrict__ X,   // (N, H, W, Cp)
+    const bf16_t* __restrict__ Wp,  // [Kp][RS*Cp]
+    const float* __restrict__ Bias,
+    bf16_t* __restrict__ Y,         // (N, H, W, Kp)
+    int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act,
+    const bf16_t* __restrict__ Zero16) {
+  static_assert(WN_MFMA_KMAP == 0, "glds staging assumes KMAP 0");
+  constexpr int BM = 256;
+  constexpr int PAD = KS / 2;
+  constexpr int RS = KS * KS;
+  constexpr int WMW = 4, WNW = 2;       // wave grid
+  constexpr int FM = BM / WMW / 16;     // 4 fragments in M per wave
+  constexpr int FN = BN / WNW / 16;     // 4 (BN=128) / 2 (BN=64)
+  constexpr int ASLOT = BM * 32 / 8 / 512;   // = 2
+  constexpr int SLOTS_B = BN * 4;
+  constexpr int BSLOT = (SLOTS_B + 511) / 512;  // = 1
+  constexpr int LA = BM * 32;           // A buffer elems
+  constexpr int LB8 = 512 * 8;          // B buffer elems (>= SLOTS_B*8)
+  const int KG = RS * Cp;
+  const long M = (long)N * H * W;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // 3 x LA
+  bf16_t* lB = lA + 3 * LA;                       // 3 x LB8
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const long m0 = (long)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int HW = H * W;
+
+  // A slots: slot = tid + s*512: i = m-frag row, kb = k-piece, mf = m-frag
+  int aOy[ASLOT], aOx[ASLOT], aKb[ASLOT];
+  long aRowBase[ASLOT];
+  bool aMv[ASLOT];
+#pragma unroll
+  for (int s = 0; s < ASLOT; ++s) {
+    int slot = tid + s * 512;
+    int i = slot & 15, kb = (slot >> 4) & 3;
+    int mf = slot >> 6;
+    long m = m0 + mf * 16 + i;
+    aMv[s] = m < M;
+    long mm = aMv[s] ? m : 0;
+    int n = (int)(mm / HW);
+    int rem = (int)(mm - (long)n * HW);
+    aOy[s] = rem / W;
+    aOx[s] = rem - aOy[s] * W;
+    aRowBase[s] = (long)n * H;
+    aKb[s] = kb;
+  }
+  int bK[BSLOT], bKb[BSLOT];
+  bool bV[BSLOT];
+#pragma unroll
+  for (int s = 0; s < BSLOT; ++s) {
+    int slot = tid + s * 512;
+    bV[s] = slot < SLOTS_B;
+    int j = slot & 15, kb = (slot >> 4) & 3;
+    int nf = slot >> 6;
+    bK[s] = n0 + nf * 16 + j;
+    bKb[s] = kb;
+  }
+
+  auto stage = [&](int buf, int ks) {
+    const int k0 = ks * 32;
+#pragma unroll
+    for (int s = 0; s < ASLOT; ++s) {
+      const bf16_t* src = Zero16;
+      int rsc = k0 + aKb[s] * 8;
+      if (aMv[s] && rsc < KG) {
+        int tap = rsc >> log2Cp;
+        int c = rsc & (Cp - 1);
+        int dy = tap / KS, dx = tap - (tap / KS) * KS;
+        int iy = aOy[s] + dy - PAD, ix = aOx[s] + dx - PAD;
+        if (iy >= 0 && iy < H && ix >= 0 && ix < W)
+          src = X + (((aRowBase[s] + iy) * W + ix) << log2Cp) + c;
+      }
+      __builtin_amdgcn_global_load_lds(
+          src, lA + buf * LA + (tid + s * 512) * 8, 16, 0, 0);
+    }
+#pragma unroll
+    for (int s = 0; s < BSLOT; ++s) {
+      const bf16_t* src = Zero16;
+      int rsc = k0 + bKb[s] * 8;
+      if (bV[s] && bK[s] < Kp && rsc < KG)
+        src = Wp + (long)bK[s] * KG + rsc;
+      __builtin_amdgcn_global_load_lds(
+          src, lB + buf * LB8 + (tid + s * 512) * 8, 16, 0, 0);
+    }
+  };
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int a = 0; a < FM; ++a)
+#pragma unroll
+    for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nk = (KG + 31) / 32;
+  const int lg = lane >> 4, li = lane & 15;
+  constexpr int GPS = ASLOT + BSLOT;  // 3 glds per thread per stage
+
+  auto wait_tiles = [&](int inflight) {
+    if (inflight > 0)
+      asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  };
+  static_assert(GPS == 3, "vmcnt immediates assume 3 glds per stage");
+
+  stage(0, 0);
+  if (1 < nk) stage(1, 1);
+  wait_tiles(1 < nk ? 1 : 0);
+  __builtin_amdgcn_s_barrier();
+
+  for (int ks = 0; ks < nk; ++ks) {
+    const int b = ks % 3;
+    if (ks + 2 < nk) stage((b + 2) % 3, ks + 2);
+    // phase 1: first half of the M fragments
+    bf16x8 aF[FM], bF[FN];
+#pragma unroll
+    for (int fn = 0; fn < FN; ++fn)
+      bF[fn] = *reinterpret_cast<const bf16x8*>(
+          lB + b * LB8 + (((wc * FN + fn) * 4 + lg) * 16 + li) * 8);
+#pragma unroll
+    for (int fm = 0; fm < FM / 2; ++fm)
+      aF[fm] = *reinterpret_cast<const bf16x8*>(
+          lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = 0; fm < FM / 2; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    // phase 2: second half
+#pragma unroll
+    for (int fm = FM / 2; fm < FM; ++fm)
+      aF[fm] = *reinterpret_cast<const bf16x8*>(
+          lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int fm = FM / 2; fm < FM; ++fm)
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    if (ks + 1 < nk) {
+      wait_tiles(ks + 2 < nk ? 1 : 0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue: bias + activation + pad zeroing
+  const int lr4 = lg * 4;
+#pragma unroll
+  for (int fn = 0; fn < FN; ++fn) {
+    const int k = n0 + (wc * FN + fn) * 16 + li;
+    if (k >= Kp) continue;
+    const float bv = (Bias != nullptr && k < Klog) ? Bias[k] : 0.f;
+    const bool kpad = k >= Klog;
+#pragma unroll
+    for (int fm = 0; fm < FM; ++fm) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long m = m0 + (wr * FM + fm) * 16 + lr4 + r;
+        if (m >= M) continue;
+        float v = acc[fm][fn][r] + bv;
+        if (act == ACT_RELU) v = fmaxf(v, 0.f);
+        else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
+        if (kpad) v = 0.f;
+        Y[m * Kp + k] = f2bf(v);
+      }
+    }
+  }
+}
+
 // Split-K finalize: Y = act(sum over gz slabs of Y32 + bias), pad zeroed.
 __global__ void k_splitk_finalize(const float* __restrict__ Y32,
                                   const float* __restrict__ Bias,
@@ -863,11 +1060,27 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
                          y32.data_ptr<float>(), bptr,
                          (bf16_t*)y.data_ptr(), M, Kp, Klog, act, gz);
     } else {
-      hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256),
-                         lds, stream, (const bf16_t*)x.data_ptr(),
-                         (const bf16_t*)wp.data_ptr(), bptr,
-                         (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,
-                         Klog, act, zptr);
+      bool done8 = false;
+      if constexpr (BN >= 64) {
+        if (M >= 16384) {  // big-M: 8-wave 256-row phase-split kernel
+          const int gx8 = (int)((M + 255) / 256);
+          const size_t lds8 =
+              (3 * 256 * 32 + 3 * 512 * 8) * sizeof(bf16_t);
+          hipLaunchKernelGGL((k_conv_igemm8<BN, KS>), dim3(gx8, gy),
+                             dim3(512), lds8, stream,
+                             (const bf16_t*)x.data_ptr(),
+                             (const bf16_t*)wp.data_ptr(), bptr,
+                             (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp),
+                             Kp, Klog, act, zptr);
+          done8 = true;
+        }
+      }
+      if (!done8)
+        hipLaunchKernelGGL((k_conv_igemm<BN, KS>), dim3(gx, gy), dim3(256),
+                           lds, stream, (const bf16_t*)x.data_ptr(),
+                           (const bf16_t*)wp.data_ptr(), bptr,
+                           (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp),
+                           Kp, Klog, act, zptr);
     }
   };
   if (Kp >= 128)
