@@ -777,6 +777,79 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 }
 
 // ---------------------------------------------------------------------------
+// Small-K weight gradient (K <= 4: the 64->3 / 32->3 output convs).
+// The MFMA path wastes 13/16 rows of every fragment on Kp padding there;
+// a VALU outer-product reduction is ~10x cheaper. Each block owns one
+// (tap, m-slice); thread t owns channel octet t % (Cp/8) and accumulates
+// acc[K][8] in registers over its m rows, LDS-reduces across the m-rows
+// sharing the octet, then atomically adds into the fp32 dW.
+// ---------------------------------------------------------------------------
+
+template <int KMAX>
+__global__ __launch_bounds__(256, 4) void k_wgrad_smallk(
+    const bf16_t* __restrict__ dY,  // (M, Kp)
+    const bf16_t* __restrict__ X,   // (N,H,W,Cp)
+    float* __restrict__ dW,         // (K, C, KS, KS) fp32
+    int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C, int KS,
+    int msplit, unsigned long long mulHW, unsigned long long mulW) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = reinterpret_cast<float*>(smem);  // [256][KMAX*8]
+  const int PAD = KS / 2;
+  const long M = (long)N * H * W;
+  const int HW = H * W;
+  const int cg = threadIdx.x & (Cp / 8 - 1);   // channel octet
+  const int mr = threadIdx.x / (Cp / 8);       // m-row lane
+  const int MR = 256 / (Cp / 8);
+  const int tap = blockIdx.x;
+  const int dy_ = tap / KS, dx_ = tap - dy_ * KS;
+
+  float acc[KMAX][8] = {};
+  const long rows = (M + msplit - 1) / msplit;
+  const long mstart = (long)blockIdx.y * rows;
+  const long mend = min(mstart + rows, M);
+  for (long m = mstart + mr; m < mend; m += MR) {
+    unsigned n = magic_div((unsigned)m, mulHW);
+    unsigned rem = (unsigned)m - n * (unsigned)HW;
+    unsigned oy = magic_div(rem, mulW);
+    int ox = (int)(rem - oy * (unsigned)W);
+    int iy = (int)oy + dy_ - PAD, ix = ox + dx_ - PAD;
+    if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+    const bf16x8 xv = *reinterpret_cast<const bf16x8*>(
+        X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + cg * 8);
+    const bf16_t* dyp = dY + m * Kp;
+#pragma unroll
+    for (int k = 0; k < KMAX; ++k) {
+      const float d = bf2f(dyp[k]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[k][e] += d * bf2f(xv[e]);
+    }
+  }
+#pragma unroll
+  for (int k = 0; k < KMAX; ++k)
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      red[(threadIdx.x) * (KMAX * 8) + k * 8 + e] = acc[k][e];
+  __syncthreads();
+  if (mr == 0) {
+#pragma unroll 1
+    for (int r = 1; r < MR; ++r)
+#pragma unroll
+      for (int k = 0; k < KMAX; ++k)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[k][e] += red[(r * (Cp / 8) + cg) * (KMAX * 8) + k * 8 + e];
+    for (int k = 0; k < K && k < KMAX; ++k)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int c = cg * 8 + e;
+        if (c < C)
+          atomicAdd(&dW[(((long)k * C + c) * KS + dy_) * KS + dx_],
+                    acc[k][e]);
+      }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // tr-read semantics probe: one wave fills a tr image with addr-coded values
 // via the SAME writeTiles addressing (tr_addr), tr-reads fragments the same
 // way wgrad does, and writes what each (lane, frag, elem) received. The
@@ -1135,6 +1208,22 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   const int Kp = dy.size(3);
   const int K = dw.size(0), C = dw.size(1);
   TORCH_CHECK(dw.size(2) == ks && dw.size(3) == ks);
+  hipStream_t stream0 = at::cuda::getCurrentHIPStream();
+  if (K <= 4) {  // output convs (K=3): VALU outer-product path
+    const int RS = (int)(ks * ks);
+    int msplit = std::max(1, 768 / RS);
+    const int MR = 256 / (Cp / 8);
+    msplit = (int)std::min<long>(msplit, ((long)N * H * W + MR - 1) / MR);
+    hipLaunchKernelGGL((k_wgrad_smallk<4>), dim3(RS, msplit), dim3(256),
+                       256 * 32 * sizeof(float), stream0,
+                       (const bf16_t*)dy.data_ptr(),
+                       (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
+                       H, W, Cp, log2i(Cp), Kp, K, C, (int)ks, msplit,
+                       MagicDiv::make((unsigned)(H * W)).mul,
+                       MagicDiv::make((unsigned)W).mul);
+    HIP_CHECK_LAST();
+    return;
+  }
   const int KG = (int)(ks * ks) * Cp;
   const int BK = std::min(Kp, 128);
   constexpr int WG_CH = 64;  // m rows per chunk (32 measured within noise
