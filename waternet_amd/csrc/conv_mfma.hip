@@ -2,31 +2,31 @@
 //
 // Replaces the reference's implicit cuDNN conv launches (SURVEY §2.2
 // K2-K9, K12-K14, K17) with hand-written NHWC bf16 kernels:
-//   - k_conv_igemm<BN,KS>: stride-1 same-pad conv fwd as implicit GEMM
-//     (M = N*H*W rows, N = output channels, K = R*S*Cp) on
-//     v_mfma_f32_16x16x32_bf16 with fused bias + ReLU/Sigmoid epilogue.
-//     Also runs dgrad (conv of dY with rotated/transposed weights packed by
-//     k_pack_dgrad).
-//   - k_conv_wgrad<KS>: weight gradient as reduction GEMM over M with
-//     fp32 atomic accumulation straight into the NCHW fp32 grad tensor.
-//   - k_bias_grad: column reduction of dY.
-//   - k_pack_fwd / k_pack_dgrad: NCHW fp32 master weights -> bf16 packed
-//     [Kp][R*S*Cp] (k-major, rsc contiguous) layouts.
+//   - k_conv_igemm<BN,KS,SPLITK>: stride-1 same-pad conv fwd as implicit
+//     GEMM (M = N*H*W rows, N = output channels, K = R*S*Cp) on
+//     v_mfma_f32_16x16x32_bf16 with fused bias + ReLU/Sigmoid epilogue and
+//     global_load_lds staging; SPLITK slices the K loop over gridDim.z into
+//     fp32 slabs (3-deep LDS ring, counted vmcnt) for small-M shapes.
+//     Also runs dgrad (conv of dY with rotated/transposed packed weights).
+//   - k_conv_igemm8<BN,KS>: 8-wave 256-row phase-split variant for big-M
+//     shapes (setprio-wrapped MFMA phases, counted vmcnt).
+//   - k_conv_wgrad<KS,BK,CH>: weight gradient as reduction GEMM over M,
+//     m-major LDS images + ds_read_b64_tr_b16 hardware transpose reads,
+//     fp32 atomic accumulation into the NCHW fp32 grad tensor.
+//   - k_wgrad_smallk: VALU outer-product path for K <= 4 output convs.
+//   - k_bias_grad(+_reduce): slab-partial column sums of dY.
+//   - k_pack_fwd / k_pack_dgrad / k_pack_all: NCHW fp32 masters -> bf16
+//     packed [Kp][R*S*Cp] (fwd) and rotated [Cp][R*S*Kp] (dgrad) layouts.
 //
 // Design notes (MI355X):
-//   - LDS tiles are stored FRAGMENT-MAJOR: slot (mf, kb, i) holds the 8
-//     bf16 the MFMA lane (kb,i) consumes, so both ds_write_b128 staging and
-//     ds_read_b128 fragment loads are bank-conflict-free without swizzles
-//     (lane l covers dwords 4*(l&15)..+3 of a 1 KiB block; the b128 lane
-//     groups each touch all 64 banks exactly once).
+//   - igemm LDS tiles are FRAGMENT-MAJOR: slot (mf, kb, i) holds the 8
+//     bf16 the MFMA lane (kb,i) consumes, so the lane-linear glds image and
+//     ds_read_b128 fragment loads are bank-conflict-free without swizzles.
 //   - Cp (physical channels) is a power of two >= 16; pad channels are
 //     zero by construction everywhere, so no channel masking in the GEMM.
-//   - Double-buffered LDS, global loads for step k+1 issued before the
-//     MFMA of step k (async-STAGE split, write-late).
-//   - 256 threads = 4 waves; BM=128 rows; BN in {16,32,64,128} columns.
 //
-// Fragment K mapping: set by WN_MFMA_KMAP (probe_mfma.hip measures it on
-// hardware). 0: lane group g holds k = g*8+e. 1: k = g*4+(e&3)+(e>>2)*16.
+// WN_MFMA_KMAP: fragment K mapping, measured by probe_mfma.hip — 0 on this
+// hardware (lane group g consumes reduction elements k = g*8+e).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -36,14 +36,6 @@
 #ifndef WN_MFMA_KMAP
 #define WN_MFMA_KMAP 0
 #endif
-
-WN_DEVFN constexpr int frag_k16(int g, int e) {
-#if WN_MFMA_KMAP == 0
-  return g * 8 + e;
-#else
-  return g * 4 + (e & 3) + ((e >> 2) << 4);
-#endif
-}
 
 // ---------------------------------------------------------------------------
 // Forward / dgrad implicit GEMM
