@@ -123,6 +123,10 @@ def main():
         use_graph=(not args.no_graph) and world == 1,
     )
 
+    if world > 1:
+        # all ranks start from rank0's params (one-time, untimed)
+        torch.distributed.broadcast(trainer.opt.master, src=0)
+
     for _ in range(max(args.warmup, 1)):
         trainer.step()
 

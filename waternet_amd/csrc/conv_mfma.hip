@@ -65,8 +65,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // SPLITK uses a 3-deep buffer ring (see below), the 2-phase path 2.
-  constexpr int NBUF = SPLITK ? 3 : 2;
+  // SPLITK uses a 5-deep buffer ring (see below), the 2-phase path 2.
+  constexpr int NBUF = SPLITK ? 5 : 2;
   bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // NBUF x BM*32
   bf16_t* lB = lA + NBUF * BM * 32;                          // NBUF x LB
 
@@ -171,23 +171,31 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     // measured NEGATIVE at the big-M shapes' 5-blocks/CU occupancy, which
     // keep the simpler 2-phase below).
     constexpr int GPS = 2 + NBS;  // glds per stage per thread (3 or 4)
-    auto wait_tiles = [&](int inflight) {
-      if (inflight > 0) {
-        if constexpr (GPS == 3)
-          asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
-        else
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    // wait until at most `tiles`' worth of glds remain outstanding
+    auto wait_tiles = [&](int tiles) {
+      if constexpr (GPS == 3) {
+        switch (tiles) {
+          case 3: asm volatile("s_waitcnt vmcnt(9)" ::: "memory"); break;
+          case 2: asm volatile("s_waitcnt vmcnt(6)" ::: "memory"); break;
+          case 1: asm volatile("s_waitcnt vmcnt(3)" ::: "memory"); break;
+          default: asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
       } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        switch (tiles) {
+          case 3: asm volatile("s_waitcnt vmcnt(12)" ::: "memory"); break;
+          case 2: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+          case 1: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+          default: asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
       }
     };
-    stage(0, ks0 * 32);
-    if (ks0 + 1 < nk) stage(1, (ks0 + 1) * 32);
-    wait_tiles(ks0 + 1 < nk ? 1 : 0);  // tile ks0 landed
+    // prologue: up to 4 tiles in flight, wait for tile ks0 only
+    for (int t = 0; t < 4 && ks0 + t < nk; ++t) stage(t, (ks0 + t) * 32);
+    wait_tiles(min(nk - ks0 - 1, 3));
     __builtin_amdgcn_s_barrier();
     for (int ks = ks0; ks < nk; ++ks) {
-      const int b = (ks - ks0) % 3;
-      if (ks + 2 < nk) stage((b + 2) % 3, (ks + 2) * 32);
+      const int b = (ks - ks0) % 5;
+      if (ks + 4 < nk) stage((b + 4) % 5, (ks + 4) * 32);
       bf16x8 aF[FM], bF[FN];
 #pragma unroll
       for (int fm = 0; fm < FM; ++fm)
@@ -204,7 +212,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
       if (ks + 1 < nk) {
-        wait_tiles(ks + 2 < nk ? 1 : 0);  // tile ks+1 landed
+        wait_tiles(min(nk - ks - 2, 3));  // tile ks+1 landed
         __builtin_amdgcn_s_barrier();
       }
     }
@@ -1111,10 +1119,10 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     gz = std::min(gz, 8);  // finalize reads gz slabs; >8 is latency-negative
     if (gz > 1) {
       auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
-      const size_t lds3 =
-          (3 * 128 * 32 + 3 * NBS * 256 * 8) * sizeof(bf16_t);
+      const size_t lds5 =
+          (5 * 128 * 32 + 5 * NBS * 256 * 8) * sizeof(bf16_t);
       hipLaunchKernelGGL((k_conv_igemm<BN, KS, true>), dim3(gx, gy, gz),
-                         dim3(256), lds3, stream,
+                         dim3(256), lds5, stream,
                          (const bf16_t*)x.data_ptr(),
                          (const bf16_t*)wp.data_ptr(), bptr,
                          (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp), Kp,

@@ -36,7 +36,8 @@ class BenchTrainer:
 
         torch.manual_seed(seed)
         self.model = WaterNet().to(self.device)
-        self.vgg = PerceptualModel(seed=seed).to(self.device).eval()
+        # common seed: the frozen VGG must be identical on every DDP rank
+        self.vgg = PerceptualModel(seed=1234).to(self.device).eval()
         self.opt = FusedAdam(self.model.parameters(), lr=lr, model=self.model)
         self.sched = torch.optim.lr_scheduler.StepLR(self.opt,
                                                      step_size=10000,
