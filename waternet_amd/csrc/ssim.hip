@@ -20,10 +20,11 @@ __constant__ float SSIM_W[SSIM_K * SSIM_K];
 
 __global__ void k_ssim(const float* __restrict__ A,
                        const float* __restrict__ B,
-                       double* __restrict__ out, int H, int W, int OH,
+                       float* __restrict__ partials, int H, int W, int OH,
                        int OW, float c1, float c2) {
   __shared__ __attribute__((aligned(16))) float sA[SSIM_PATCH][SSIM_PATCH];
   __shared__ __attribute__((aligned(16))) float sB[SSIM_PATCH][SSIM_PATCH];
+  __shared__ float red[4];
 
   const int plane = blockIdx.z;  // n*C + c
   const float* a = A + (long)plane * H * W;
@@ -67,9 +68,16 @@ __global__ void k_ssim(const float* __restrict__ A,
     float den = (sx * sx + sy * sy + c1) * (vx + vy + c2);
     ssim = num / den;
   }
-  // wave reduce + one atomic per wave
+  // wave reduce -> LDS -> ONE fp32 partial per block (a single f64
+  // accumulator serialized ~9.4K atomics and dominated the kernel)
   for (int off = 32; off > 0; off >>= 1) ssim += __shfl_down(ssim, off, 64);
-  if ((threadIdx.x & 63) == 0) atomicAdd(out, (double)ssim);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ssim;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int bid =
+        (blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x + blockIdx.x;
+    partials[bid] = red[0] + red[1] + red[2] + red[3];
+  }
 }
 
 at::Tensor ssim_sum(const at::Tensor& a, const at::Tensor& b,
@@ -94,15 +102,16 @@ at::Tensor ssim_sum(const at::Tensor& a, const at::Tensor& b,
   const int N = a.size(0), C = a.size(1), H = a.size(2), W = a.size(3);
   const int OH = H - SSIM_K + 1, OW = W - SSIM_K + 1;
   TORCH_CHECK(OH > 0 && OW > 0, "image smaller than SSIM window");
-  auto out = at::zeros({}, a.options().dtype(at::kDouble));
   const float c1 = (float)((k1 * data_range) * (k1 * data_range));
   const float c2 = (float)((k2 * data_range) * (k2 * data_range));
   dim3 grid((OW + 15) / 16, (OH + 15) / 16, N * C);
+  auto partials = at::empty({(long)grid.x * grid.y * grid.z},
+                            a.options().dtype(at::kFloat));
   hipStream_t stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(k_ssim, grid, dim3(256), 0, stream,
                      a.contiguous().data_ptr<float>(),
                      b.contiguous().data_ptr<float>(),
-                     out.data_ptr<double>(), H, W, OH, OW, c1, c2);
+                     partials.data_ptr<float>(), H, W, OH, OW, c1, c2);
   HIP_CHECK_LAST();
-  return out;  // caller divides by N*C*OH*OW
+  return partials.sum(at::kDouble);  // caller divides by N*C*OH*OW
 }
