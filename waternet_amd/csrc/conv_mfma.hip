@@ -837,21 +837,27 @@ __global__ __launch_bounds__(256, 4) void k_wgrad_smallk(
     for (int e = 0; e < 8; ++e)
       red[(threadIdx.x) * (KMAX * 8) + k * 8 + e] = acc[k][e];
   __syncthreads();
-  if (mr == 0) {
-#pragma unroll 1
-    for (int r = 1; r < MR; ++r)
+  // parallel tree reduce over the MR m-rows sharing each channel octet
+  const int CG = Cp / 8;
+  for (int s = MR / 2; s > 0; s >>= 1) {
+    if (mr < s) {
 #pragma unroll
       for (int k = 0; k < KMAX; ++k)
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          acc[k][e] += red[(r * (Cp / 8) + cg) * (KMAX * 8) + k * 8 + e];
+          red[(mr * CG + cg) * (KMAX * 8) + k * 8 + e] +=
+              red[((mr + s) * CG + cg) * (KMAX * 8) + k * 8 + e];
+    }
+    __syncthreads();
+  }
+  if (mr == 0) {
     for (int k = 0; k < K && k < KMAX; ++k)
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const int c = cg * 8 + e;
         if (c < C)
           atomicAdd(&dW[(((long)k * C + c) * KS + dy_) * KS + dx_],
-                    acc[k][e]);
+                    red[cg * (KMAX * 8) + k * 8 + e]);
       }
   }
 }
