@@ -1251,8 +1251,10 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   }
   const int KG = (int)(ks * ks) * Cp;
   const int BK = std::min(Kp, 128);
-  constexpr int WG_CH = 64;  // m rows per chunk (32 measured within noise
-                             // of 64 at bs=16 112^2 despite 2x occupancy)
+  // m rows per chunk: 64 for the wide tiles; 128 for BK<=32 (those do only
+  // 8 MFMAs per wave between barriers at CH=64 — amortize the chunk
+  // overhead; LDS still fits 3 blocks/CU).
+  const int WG_CH = (BK <= 32) ? 128 : 64;
   const int gx = (Kp + BK - 1) / BK, gy = (KG + 127) / 128;
   // split so gx*gy*split fills 256 CUs x ~3 resident blocks
   int split = std::max(1, 640 / std::max(1, gx * gy));
@@ -1264,7 +1266,8 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
-    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, WG_CH>), dim3(gx, gy, split),
+    constexpr int CHV = (BKV <= 32) ? 128 : 64;
+    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, CHV>), dim3(gx, gy, split),
                        dim3(256), lds, stream, (const bf16_t*)dy.data_ptr(),
                        (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
                        H, W, Cp, log2i(Cp), Kp, K, C, split,
