@@ -140,3 +140,21 @@ def test_histeq_only_touches_luminance():
     # red stays dominant
     assert he[:, :, 0].mean() > he[:, :, 1].mean()
     assert he[:, :, 0].mean() > he[:, :, 2].mean()
+
+
+def test_white_balance_grayscale_path():
+    """Reference data.py:32-44: grayscale input uses satLevels
+    [0.001, 0.005] and must not mutate the input (the reference's in-place
+    reshape-view mutation is a documented latent bug we do not replicate)."""
+    import numpy as np
+
+    from waternet_amd.data.transforms import white_balance_transform
+
+    rng = np.random.default_rng(3)
+    gray = rng.integers(20, 200, size=(32, 24), dtype=np.uint8)
+    before = gray.copy()
+    out = white_balance_transform(gray)
+    assert out.shape == gray.shape and out.dtype == np.uint8
+    assert np.array_equal(gray, before), "input mutated"
+    # stretch reaches (close to) the full range
+    assert out.max() >= 250 and out.min() <= 5
