@@ -291,7 +291,10 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
 //   - two MFMA phases per K-step with s_setprio(1) around the matrix math.
 // ---------------------------------------------------------------------------
 
-template <int BN, int KS>
+// VAR: 0 = 2 setprio phases / 3-ring (default); 1 = no setprio;
+// 2 = 4 phases; 3 = 4-deep ring (2 tiles in flight). Runtime-selectable
+// via WN_IGEMM8_VAR for within-box A/B (guide §5.4 rule 24).
+template <int BN, int KS, int VAR = 0>
 __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
     const bf16_t* __restrict__ X,   // (N, H, W, Cp)
     const bf16_t* __restrict__ Wp,  // [Kp][RS*Cp]
@@ -395,60 +398,53 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
   const int nk = (KG + 31) / 32;
   const int lg = lane >> 4, li = lane & 15;
   constexpr int GPS = ASLOT + BSLOT;  // 3 glds per thread per stage
+  constexpr int RING = (VAR == 3) ? 4 : 3;
+  constexpr int AHEAD = RING - 2;  // tiles left in flight at the wait
+  constexpr bool PRIO = (VAR != 1);
+  constexpr int NPH = (VAR == 2) ? 4 : 2;  // M-fragment phases per K-step
 
   auto wait_tiles = [&](int inflight) {
-    if (inflight > 0)
+    if (inflight >= 2 && AHEAD == 2)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if (inflight >= 1)
       asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   };
   static_assert(GPS == 3, "vmcnt immediates assume 3 glds per stage");
 
-  stage(0, 0);
-  if (1 < nk) stage(1, 1);
-  wait_tiles(1 < nk ? 1 : 0);
+  for (int t = 0; t < RING - 1 && t < nk; ++t) stage(t, t);
+  wait_tiles(min(nk - 1, RING - 2));
   __builtin_amdgcn_s_barrier();
 
   for (int ks = 0; ks < nk; ++ks) {
-    const int b = ks % 3;
-    if (ks + 2 < nk) stage((b + 2) % 3, ks + 2);
-    // phase 1: first half of the M fragments
+    const int b = ks % RING;
+    if (ks + RING - 1 < nk) stage((b + RING - 1) % RING, ks + RING - 1);
     bf16x8 aF[FM], bF[FN];
 #pragma unroll
     for (int fn = 0; fn < FN; ++fn)
       bF[fn] = *reinterpret_cast<const bf16x8*>(
           lB + b * LB8 + (((wc * FN + fn) * 4 + lg) * 16 + li) * 8);
 #pragma unroll
-    for (int fm = 0; fm < FM / 2; ++fm)
-      aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
-    __builtin_amdgcn_s_setprio(1);
+    for (int ph = 0; ph < NPH; ++ph) {
+      constexpr int FPP = FM / NPH;  // fragments per phase
 #pragma unroll
-    for (int fm = 0; fm < FM / 2; ++fm)
+      for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
+        aF[fm] = *reinterpret_cast<const bf16x8*>(
+            lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int fn = 0; fn < FN; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    // phase 2: second half
+      for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
 #pragma unroll
-    for (int fm = FM / 2; fm < FM; ++fm)
-      aF[fm] = *reinterpret_cast<const bf16x8*>(
-          lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int fm = FM / 2; fm < FM; ++fm)
-#pragma unroll
-      for (int fn = 0; fn < FN; ++fn)
-        acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
+        for (int fn = 0; fn < FN; ++fn)
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+    }
     if (ks + 1 < nk) {
-      wait_tiles(ks + 2 < nk ? 1 : 0);
+      wait_tiles(min(nk - ks - 2, AHEAD));
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -1161,15 +1157,29 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
       bool done8 = false;
       if constexpr (BN >= 64) {
         if (M >= 16384) {  // big-M: 8-wave 256-row phase-split kernel
+          static const int var = [] {
+            const char* e = getenv("WN_IGEMM8_VAR");
+            return e ? atoi(e) : 0;
+          }();
           const int gx8 = (int)((M + 255) / 256);
+          const int ring = (var == 3) ? 4 : 3;
           const size_t lds8 =
-              (3 * 256 * 32 + 3 * 512 * 8) * sizeof(bf16_t);
-          hipLaunchKernelGGL((k_conv_igemm8<BN, KS>), dim3(gx8, gy),
-                             dim3(512), lds8, stream,
-                             (const bf16_t*)x.data_ptr(),
-                             (const bf16_t*)wp.data_ptr(), bptr,
-                             (bf16_t*)y.data_ptr(), N, H, W, Cp, log2i(Cp),
-                             Kp, Klog, act, zptr);
+              (size_t)(ring * 256 * 32 + ring * 512 * 8) * sizeof(bf16_t);
+          auto l8 = [&](auto var_const) {
+            constexpr int V = decltype(var_const)::value;
+            hipLaunchKernelGGL((k_conv_igemm8<BN, KS, V>), dim3(gx8, gy),
+                               dim3(512), lds8, stream,
+                               (const bf16_t*)x.data_ptr(),
+                               (const bf16_t*)wp.data_ptr(), bptr,
+                               (bf16_t*)y.data_ptr(), N, H, W, Cp,
+                               log2i(Cp), Kp, Klog, act, zptr);
+          };
+          switch (var) {
+            case 1: l8(std::integral_constant<int, 1>{}); break;
+            case 2: l8(std::integral_constant<int, 2>{}); break;
+            case 3: l8(std::integral_constant<int, 3>{}); break;
+            default: l8(std::integral_constant<int, 0>{});
+          }
           done8 = true;
         }
       }
