@@ -85,10 +85,22 @@ class PerceptualModel(nn.Module):
         return unexpected
 
 
+_NORM_CACHE = {}
+
+
 def normalize_imagenet(x: torch.Tensor) -> torch.Tensor:
-    """TF.normalize(x, ImageNet mean/std) equivalent (train.py:111-116)."""
-    mean = x.new_tensor(IMAGENET_MEAN).view(1, 3, 1, 1)
-    std = x.new_tensor(IMAGENET_STD).view(1, 3, 1, 1)
+    """TF.normalize(x, ImageNet mean/std) equivalent (train.py:111-116).
+
+    The mean/std constants are cached per (device, dtype): building them
+    with new_tensor() is a pageable H2D copy, which is illegal inside
+    hipGraph capture (it silently broke whole-step capture)."""
+    key = (x.device, x.dtype)
+    cached = _NORM_CACHE.get(key)
+    if cached is None:
+        mean = x.new_tensor(IMAGENET_MEAN).view(1, 3, 1, 1)
+        std = x.new_tensor(IMAGENET_STD).view(1, 3, 1, 1)
+        cached = _NORM_CACHE[key] = (mean, std)
+    mean, std = cached
     return (x - mean) / std
 
 
