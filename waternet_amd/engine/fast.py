@@ -170,6 +170,16 @@ class BenchTrainer:
     def _maybe_capture(self):
         if self._graph is not None or not self._use_graph:
             return
+        # The 2 warmup executions advance the optimizer/metric state; keep
+        # capture side-effect-free by snapshotting and restoring it so the
+        # first replay computes exactly the step an eager run would.
+        o = self.opt
+        snap = {
+            "master": o.master.clone(), "grads": o.grads.clone(),
+            "exp_avg": o.exp_avg.clone(), "exp_avg_sq": o.exp_avg_sq.clone(),
+            "step_buf": o.step_buf.clone(), "metric": self.metric_sums.clone(),
+        }
+        host_step = o._step
         # warmup on a side stream, then capture one step
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -188,6 +198,15 @@ class BenchTrainer:
             print(f"[fast] hipGraph capture failed, running eager: {e!r}",
                   file=sys.stderr)
             self._use_graph = False
+        # restore pre-warmup state (capture itself records, not executes)
+        o.master.copy_(snap["master"])
+        o.grads.copy_(snap["grads"])
+        o.exp_avg.copy_(snap["exp_avg"])
+        o.exp_avg_sq.copy_(snap["exp_avg_sq"])
+        o.step_buf.copy_(snap["step_buf"])
+        self.metric_sums.copy_(snap["metric"])
+        o._step = host_step
+        o._mark_model_dirty()  # packed-weight caches reflect restored masters
 
     def step(self):
         self._load_batch()
