@@ -577,7 +577,7 @@ WN_DEVFN bf16x4 ds_tr16(unsigned addr) {
   return v;
 }
 
-template <int KS, int BK, int CH = 64>  // CH: m rows per chunk (32 or 64)
+template <int KS, int BK, int CH = 64, bool M32 = false>
 __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     const bf16_t* __restrict__ dY,  // (N,H,W,Kp)
     const bf16_t* __restrict__ X,   // (N,H,W,Cp)
@@ -690,11 +690,25 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     }
   };
 
-  f32x4 acc[FK][FR];
+  // M32 variant: v_mfma_f32_32x32x16_bf16 — half the MFMA instruction
+  // count for the same MACs at the 2382-vs-2075 TF pipe rate (PMC shows
+  // the 16x16 wgrad is MFMA-pipe-saturated). Wave tile stays 64x64
+  // (BK>=64): FK32 x FR32 accumulators of 16 f32.
+  constexpr int FK32 = (BK >= 64) ? BK / WR / 32 : 1;
+  constexpr int FR32 = BR / WC / 32;
+  f32x4 acc[M32 ? 1 : FK][M32 ? 1 : FR];
+  f32x16 acc32[M32 ? FK32 : 1][M32 ? FR32 : 1];
 #pragma unroll
-  for (int a = 0; a < FK; ++a)
+  for (int a = 0; a < (M32 ? FK32 : FK); ++a)
 #pragma unroll
-    for (int b = 0; b < FR; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < (M32 ? FR32 : FR); ++b) {
+      if constexpr (M32) {
+#pragma unroll
+        for (int e = 0; e < 16; ++e) acc32[a][b][e] = 0.f;
+      } else {
+        acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+      }
+    }
 
   const long nChunks = (M + CH - 1) / CH;
   const int lg = lane >> 4, li = lane & 15;
@@ -704,6 +718,13 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
                         (unsigned)(2 * lg * TR_MBS + li * 4) * 2;
   const unsigned bTr0 = (unsigned)(unsigned long long)lB +
                         (unsigned)(2 * lg * TR_MBS + li * 4) * 2;
+  // M32 per-lane bases: lane l reads column (l&31) => col-block (l>>4)&1,
+  // reduction m = (l>>5)*8 + e => mblk offset (l>>5)*2 (+h)
+  const unsigned trM32 =
+      (unsigned)(((lane >> 4) & 1) * KBS + (lane >> 5) * 2 * TR_MBS +
+                 (lane & 15) * 4) * 2;
+  const unsigned aTrM = (unsigned)(unsigned long long)lA + trM32;
+  const unsigned bTrM = (unsigned)(unsigned long long)lB + trM32;
 
   long chunk = blockIdx.z;
   if (chunk < nChunks) {
@@ -718,43 +739,118 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
       loadA(next * CH);  // global -> regs, overlaps this chunk's math
       loadB(next * CH);
     }
-    // tr-read ALL fragments for this chunk into registers (guide T10)
-    bf16x4 aT[CH / 32][FK][2], bT[CH / 32][FR][2];
+    if constexpr (M32) {
+      // two half-chunk batches (2 x 16-m ksteps each) keep the tr-read
+      // register footprint at 16 bf16x4
+      constexpr int KST = CH / 16;  // 32x32x16 ksteps per chunk
 #pragma unroll
-    for (int s2 = 0; s2 < CH / 32; ++s2) {
+      for (int hb = 0; hb < 2; ++hb) {
+        bf16x4 aT[KST / 2][FK32][2], bT[KST / 2][FR32][2];
 #pragma unroll
-      for (int f = 0; f < FK; ++f) {
-        const unsigned base = aTr0 + (unsigned)((wr * FK + f) * KBS +
-                                                s2 * 8 * TR_MBS) * 2;
-        aT[s2][f][0] = ds_tr16(base);
-        aT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+        for (int s = 0; s < KST / 2; ++s) {
+          const int kst = hb * (KST / 2) + s;
+#pragma unroll
+          for (int f = 0; f < FK32; ++f) {
+            const unsigned base =
+                aTrM + (unsigned)((wr * (BK / WR / 16) + f * 2) * KBS +
+                                  kst * 4 * TR_MBS) * 2;
+            aT[s][f][0] = ds_tr16(base);
+            aT[s][f][1] = ds_tr16(base + TR_MBS * 2);
+          }
+#pragma unroll
+          for (int f = 0; f < FR32; ++f) {
+            const unsigned base =
+                bTrM + (unsigned)((wc * (BR / WC / 16) + f * 2) * KBS +
+                                  kst * 4 * TR_MBS) * 2;
+            bT[s][f][0] = ds_tr16(base);
+            bT[s][f][1] = ds_tr16(base + TR_MBS * 2);
+          }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        if (hb == 1) {
+          __syncthreads();  // ALL reads done; LDS free for restaging
+          if (next < nChunks) writeTiles();
+        }
+#pragma unroll
+        for (int s = 0; s < KST / 2; ++s)
+#pragma unroll
+          for (int fa = 0; fa < FK32; ++fa)
+#pragma unroll
+            for (int fb = 0; fb < FR32; ++fb) {
+              const bf16x8 av = __builtin_shufflevector(
+                  aT[s][fa][0], aT[s][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
+              const bf16x8 bv = __builtin_shufflevector(
+                  bT[s][fb][0], bT[s][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
+              acc32[fa][fb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  av, bv, acc32[fa][fb], 0, 0, 0);
+            }
       }
+    } else {
+      // tr-read ALL fragments for this chunk into registers (guide T10)
+      bf16x4 aT[CH / 32][FK][2], bT[CH / 32][FR][2];
 #pragma unroll
-      for (int f = 0; f < FR; ++f) {
-        const unsigned base = bTr0 + (unsigned)((wc * FR + f) * KBS +
-                                                s2 * 8 * TR_MBS) * 2;
-        bT[s2][f][0] = ds_tr16(base);
-        bT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+      for (int s2 = 0; s2 < CH / 32; ++s2) {
+#pragma unroll
+        for (int f = 0; f < FK; ++f) {
+          const unsigned base = aTr0 + (unsigned)((wr * FK + f) * KBS +
+                                                  s2 * 8 * TR_MBS) * 2;
+          aT[s2][f][0] = ds_tr16(base);
+          aT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+        }
+#pragma unroll
+        for (int f = 0; f < FR; ++f) {
+          const unsigned base = bTr0 + (unsigned)((wc * FR + f) * KBS +
+                                                  s2 * 8 * TR_MBS) * 2;
+          bT[s2][f][0] = ds_tr16(base);
+          bT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);  // keep MFMAs below the wait
+      __syncthreads();  // all waves done reading; LDS free for restaging
+      if (next < nChunks) writeTiles();  // issues alongside the MFMAs below
+#pragma unroll
+      for (int s2 = 0; s2 < CH / 32; ++s2)
+#pragma unroll
+        for (int fa = 0; fa < FK; ++fa)
+#pragma unroll
+          for (int fb = 0; fb < FR; ++fb) {
+            const bf16x8 av = __builtin_shufflevector(
+                aT[s2][fa][0], aT[s2][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
+            const bf16x8 bv = __builtin_shufflevector(
+                bT[s2][fb][0], bT[s2][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
+            acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                av, bv, acc[fa][fb], 0, 0, 0);
+          }
+    }
+    if (next < nChunks) __syncthreads();  // writes visible before next reads
+  }
+
+  if constexpr (M32) {
+    // ---- 32x32 epilogue: D col = lane&31 (rsc), row = (r&3) + 8*(r>>2)
+    //      + 4*(lane>>5) (k) ----
+#pragma unroll
+    for (int fa = 0; fa < FK32; ++fa) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        const int k = kt0 + wr * (BK / WR) + fa * 32 + row;
+        if (k >= K) continue;
+#pragma unroll
+        for (int fb = 0; fb < FR32; ++fb) {
+          const int rsc = rt0 + wc * (BR / WC) + fb * 32 + (lane & 31);
+          if (rsc >= KG) continue;
+          const int tap = rsc >> log2Cp;
+          const int c = rsc & (Cp - 1);
+          if (c >= C) continue;
+          const int dy_ = tap / KS, dx_ = tap - (tap / KS) * KS;
+          atomicAdd(&dW[(((long)k * C + c) * KS + dy_) * KS + dx_],
+                    acc32[fa][fb][r]);
+        }
       }
     }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    __builtin_amdgcn_sched_barrier(0);  // keep MFMAs below the wait
-    __syncthreads();  // all waves done reading; LDS free for restaging
-    if (next < nChunks) writeTiles();  // issues alongside the MFMAs below
-#pragma unroll
-    for (int s2 = 0; s2 < CH / 32; ++s2)
-#pragma unroll
-      for (int fa = 0; fa < FK; ++fa)
-#pragma unroll
-        for (int fb = 0; fb < FR; ++fb) {
-          const bf16x8 av = __builtin_shufflevector(
-              aT[s2][fa][0], aT[s2][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
-          const bf16x8 bv = __builtin_shufflevector(
-              bT[s2][fb][0], bT[s2][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
-          acc[fa][fb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              av, bv, acc[fa][fb], 0, 0, 0);
-        }
-    if (next < nChunks) __syncthreads();  // writes visible before next reads
+    return;
   }
 
   // ---- epilogue: scatter-add into NCHW fp32 dW ----
@@ -1273,16 +1369,28 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   hipStream_t stream = at::cuda::getCurrentHIPStream();
   const size_t lds =
       (size_t)(BK / 16 + 128 / 16) * (WG_CH / 4 * 72 + 16) * sizeof(bf16_t);
+  static const bool use_m32 = [] {
+    const char* e = getenv("WN_WGRAD_M32");
+    return e == nullptr || atoi(e) != 0;  // default ON; 0 = 16x16 path
+  }();
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
     constexpr int CHV = (BKV <= 32) ? 128 : 64;
-    hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, CHV>), dim3(gx, gy, split),
-                       dim3(256), lds, stream, (const bf16_t*)dy.data_ptr(),
-                       (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(), N,
-                       H, W, Cp, log2i(Cp), Kp, K, C, split,
-                       MagicDiv::make((unsigned)(H * W)).mul,
-                       MagicDiv::make((unsigned)W).mul);
+    auto go = [&](auto m32_const) {
+      constexpr bool M32V = decltype(m32_const)::value;
+      hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, CHV, M32V>),
+                         dim3(gx, gy, split), dim3(256), lds, stream,
+                         (const bf16_t*)dy.data_ptr(),
+                         (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
+                         N, H, W, Cp, log2i(Cp), Kp, K, C, split,
+                         MagicDiv::make((unsigned)(H * W)).mul,
+                         MagicDiv::make((unsigned)W).mul);
+    };
+    if (BKV >= 32 && use_m32)
+      go(std::integral_constant<bool, (BKV >= 32)>{});
+    else
+      go(std::false_type{});
   };
   auto launch_ks = [&](auto ks_const) {
     if (BK == 128)
