@@ -294,7 +294,9 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
 // VAR: 0 = 2 setprio phases / 3-ring (default); 1 = no setprio;
 // 2 = 4 phases; 3 = 4-deep ring (2 tiles in flight). Runtime-selectable
 // via WN_IGEMM8_VAR for within-box A/B (guide §5.4 rule 24).
-template <int BN, int KS, int VAR = 0>
+// M32: v_mfma_f32_32x32x16_bf16 fragments (half the MFMA instruction
+// count at the higher 32x32 pipe rate; +5.5% proven on the wgrad).
+template <int BN, int KS, int VAR = 0, bool M32 = false>
 __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
     const bf16_t* __restrict__ X,   // (N, H, W, Cp)
     const bf16_t* __restrict__ Wp,  // [Kp][RS*Cp]
@@ -330,16 +332,26 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
   const int n0 = blockIdx.y * BN;
   const int HW = H * W;
 
-  // A slots: slot = tid + s*512: i = m-frag row, kb = k-piece, mf = m-frag
+  // A slots: the lane-linear LDS image IS the fragment layout, so the
+  // slot->(m row, k offset) decode differs per MFMA shape:
+  //   16x16x32: slot = (mf16, kb, i16) -> m = mf16*16+i16, kOff = kb*8
+  //   32x32x16: slot = (mf32, t, g, i32) -> m = mf32*32+i32,
+  //             kOff = t*16 + g*8  (lane l reads k = (l>>5)*8+e)
   int aOy[ASLOT], aOx[ASLOT], aKb[ASLOT];
   long aRowBase[ASLOT];
   bool aMv[ASLOT];
 #pragma unroll
   for (int s = 0; s < ASLOT; ++s) {
     int slot = tid + s * 512;
-    int i = slot & 15, kb = (slot >> 4) & 3;
-    int mf = slot >> 6;
-    long m = m0 + mf * 16 + i;
+    int mrow, koff;
+    if constexpr (M32) {
+      mrow = (slot >> 7) * 32 + (slot & 31);
+      koff = ((slot >> 6) & 1) * 16 + ((slot >> 5) & 1) * 8;
+    } else {
+      mrow = (slot >> 6) * 16 + (slot & 15);
+      koff = ((slot >> 4) & 3) * 8;
+    }
+    long m = m0 + mrow;
     aMv[s] = m < M;
     long mm = aMv[s] ? m : 0;
     int n = (int)(mm / HW);
@@ -347,7 +359,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
     aOy[s] = rem / W;
     aOx[s] = rem - aOy[s] * W;
     aRowBase[s] = (long)n * H;
-    aKb[s] = kb;
+    aKb[s] = koff;
   }
   int bK[BSLOT], bKb[BSLOT];
   bool bV[BSLOT];
@@ -355,10 +367,16 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
   for (int s = 0; s < BSLOT; ++s) {
     int slot = tid + s * 512;
     bV[s] = slot < SLOTS_B;
-    int j = slot & 15, kb = (slot >> 4) & 3;
-    int nf = slot >> 6;
-    bK[s] = n0 + nf * 16 + j;
-    bKb[s] = kb;
+    int col, koff;
+    if constexpr (M32) {
+      col = (slot >> 7) * 32 + (slot & 31);
+      koff = ((slot >> 6) & 1) * 16 + ((slot >> 5) & 1) * 8;
+    } else {
+      col = (slot >> 6) * 16 + (slot & 15);
+      koff = ((slot >> 4) & 3) * 8;
+    }
+    bK[s] = n0 + col;
+    bKb[s] = koff;
   }
 
   auto stage = [&](int buf, int ks) {
@@ -366,7 +384,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
 #pragma unroll
     for (int s = 0; s < ASLOT; ++s) {
       const bf16_t* src = Zero16;
-      int rsc = k0 + aKb[s] * 8;
+      int rsc = k0 + aKb[s];
       if (aMv[s] && rsc < KG) {
         int tap = rsc >> log2Cp;
         int c = rsc & (Cp - 1);
@@ -381,7 +399,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
 #pragma unroll
     for (int s = 0; s < BSLOT; ++s) {
       const bf16_t* src = Zero16;
-      int rsc = k0 + bKb[s] * 8;
+      int rsc = k0 + bKb[s];
       if (bV[s] && bK[s] < Kp && rsc < KG)
         src = Wp + (long)bK[s] * KG + rsc;
       __builtin_amdgcn_global_load_lds(
@@ -389,11 +407,21 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
     }
   };
 
-  f32x4 acc[FM][FN];
+  constexpr int FM32 = BM / WMW / 32;       // 2
+  constexpr int FN32 = (BN / WNW) / 32;     // 2 (BN=128) / 1 (BN=64)
+  f32x4 acc[M32 ? 1 : FM][M32 ? 1 : FN];
+  f32x16 acc32[M32 ? FM32 : 1][M32 ? FN32 : 1];
 #pragma unroll
-  for (int a = 0; a < FM; ++a)
+  for (int a = 0; a < (M32 ? FM32 : FM); ++a)
 #pragma unroll
-    for (int b = 0; b < FN; ++b) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < (M32 ? FN32 : FN); ++b) {
+      if constexpr (M32) {
+#pragma unroll
+        for (int e = 0; e < 16; ++e) acc32[a][b][e] = 0.f;
+      } else {
+        acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+      }
+    }
 
   const int nk = (KG + 31) / 32;
   const int lg = lane >> 4, li = lane & 15;
@@ -420,28 +448,56 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
   for (int ks = 0; ks < nk; ++ks) {
     const int b = ks % RING;
     if (ks + RING - 1 < nk) stage((b + RING - 1) % RING, ks + RING - 1);
-    bf16x8 aF[FM], bF[FN];
+    if constexpr (M32) {
+      // lane-linear fragment reads: block (mf32*2+t)*2+g at lane*16B
+      bf16x8 aF[FM32][2], bF[FN32][2];
 #pragma unroll
-    for (int fn = 0; fn < FN; ++fn)
-      bF[fn] = *reinterpret_cast<const bf16x8*>(
-          lB + b * LB8 + (((wc * FN + fn) * 4 + lg) * 16 + li) * 8);
+      for (int t = 0; t < 2; ++t)
 #pragma unroll
-    for (int ph = 0; ph < NPH; ++ph) {
-      constexpr int FPP = FM / NPH;  // fragments per phase
+        for (int fn = 0; fn < FN32; ++fn)
+          bF[fn][t] = *reinterpret_cast<const bf16x8*>(
+              lB + b * LB8 + ((wc * FN32 + fn) * 2 + t) * 512 + lane * 8);
 #pragma unroll
-      for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
-        aF[fm] = *reinterpret_cast<const bf16x8*>(
-            lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
-      if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+      for (int ph = 0; ph < 2; ++ph) {  // one M fragment per phase
 #pragma unroll
-      for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
+        for (int t = 0; t < 2; ++t)
+          aF[ph][t] = *reinterpret_cast<const bf16x8*>(
+              lA + b * LA + ((wr * FM32 + ph) * 2 + t) * 512 + lane * 8);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int fn = 0; fn < FN; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-      if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+        for (int t = 0; t < 2; ++t)
+#pragma unroll
+          for (int fn = 0; fn < FN32; ++fn)
+            acc32[ph][fn] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                aF[ph][t], bF[fn][t], acc32[ph][fn], 0, 0, 0);
+        if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      }
+    } else {
+      bf16x8 aF[FM], bF[FN];
+#pragma unroll
+      for (int fn = 0; fn < FN; ++fn)
+        bF[fn] = *reinterpret_cast<const bf16x8*>(
+            lB + b * LB8 + (((wc * FN + fn) * 4 + lg) * 16 + li) * 8);
+#pragma unroll
+      for (int ph = 0; ph < NPH; ++ph) {
+        constexpr int FPP = M32 ? 1 : FM / NPH;  // fragments per phase
+#pragma unroll
+        for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
+          aF[fm] = *reinterpret_cast<const bf16x8*>(
+              lA + b * LA + (((wr * FM + fm) * 4 + lg) * 16 + li) * 8);
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        if constexpr (PRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int fm = ph * FPP; fm < (ph + 1) * FPP; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < FN; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
+        if constexpr (PRIO) __builtin_amdgcn_s_setprio(0);
+      }
     }
     if (ks + 1 < nk) {
       wait_tiles(min(nk - ks - 2, AHEAD));
@@ -449,6 +505,31 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
     }
   }
 
+  if constexpr (M32) {
+    // 32x32 D map: col = lane&31 (k-out), row = (r&3)+8*(r>>2)+4*(lane>>5)
+#pragma unroll
+    for (int fn = 0; fn < FN32; ++fn) {
+      const int k = n0 + (wc * FN32 + fn) * 32 + (lane & 31);
+      if (k >= Kp) continue;
+      const float bv = (Bias != nullptr && k < Klog) ? Bias[k] : 0.f;
+      const bool kpad = k >= Klog;
+#pragma unroll
+      for (int fm = 0; fm < FM32; ++fm) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+          long m = m0 + (wr * FM32 + fm) * 32 + row;
+          if (m >= M) continue;
+          float v = acc32[fm][fn][r] + bv;
+          if (act == ACT_RELU) v = fmaxf(v, 0.f);
+          else if (act == ACT_SIGMOID) v = 1.f / (1.f + __expf(-v));
+          if (kpad) v = 0.f;
+          Y[m * Kp + k] = f2bf(v);
+        }
+      }
+    }
+    return;
+  }
   // epilogue: bias + activation + pad zeroing
   const int lr4 = lg * 4;
 #pragma unroll
@@ -1257,6 +1338,10 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
             const char* e = getenv("WN_IGEMM8_VAR");
             return e ? atoi(e) : 0;
           }();
+          static const bool m32 = [] {
+            const char* e = getenv("WN_IGEMM8_M32");
+            return e == nullptr || atoi(e) != 0;  // default ON
+          }();
           const int gx8 = (int)((M + 255) / 256);
           const int ring = (var == 3) ? 4 : 3;
           const size_t lds8 =
@@ -1270,11 +1355,20 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
                                (bf16_t*)y.data_ptr(), N, H, W, Cp,
                                log2i(Cp), Kp, Klog, act, zptr);
           };
-          switch (var) {
-            case 1: l8(std::integral_constant<int, 1>{}); break;
-            case 2: l8(std::integral_constant<int, 2>{}); break;
-            case 3: l8(std::integral_constant<int, 3>{}); break;
-            default: l8(std::integral_constant<int, 0>{});
+          if (m32) {
+            hipLaunchKernelGGL((k_conv_igemm8<BN, KS, 0, true>),
+                               dim3(gx8, gy), dim3(512), lds8, stream,
+                               (const bf16_t*)x.data_ptr(),
+                               (const bf16_t*)wp.data_ptr(), bptr,
+                               (bf16_t*)y.data_ptr(), N, H, W, Cp,
+                               log2i(Cp), Kp, Klog, act, zptr);
+          } else {
+            switch (var) {
+              case 1: l8(std::integral_constant<int, 1>{}); break;
+              case 2: l8(std::integral_constant<int, 2>{}); break;
+              case 3: l8(std::integral_constant<int, 3>{}); break;
+              default: l8(std::integral_constant<int, 0>{});
+            }
           }
           done8 = true;
         }
