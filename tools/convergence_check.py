@@ -1,0 +1,98 @@
+"""Quality-parity evidence: train the SAME WaterNet on the SAME synthetic
+data/seed with (a) the native CDNA4 bf16 kernel engine and (b) eager fp32
+PyTorch ops (WATERNET_AMD_EAGER path), and compare loss/SSIM/PSNR curves.
+
+The UIEB dataset is not downloadable in this offline environment, so the
+reference's published val PSNR/SSIM cannot be reproduced directly; this
+check demonstrates the engine optimizes the identical objective on the
+identical model to the plain-PyTorch reference path (plus the per-op parity
+tests in tests/test_gpu_ops.py). Writes profiles/convergence.json.
+"""
+
+import json
+import os
+import pathlib
+import sys
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
+
+import numpy as np
+import torch
+
+from waternet_amd.engine.losses import composite_loss
+from waternet_amd.models.vgg import PerceptualModel
+from waternet_amd.models.waternet import WaterNet
+from waternet_amd.ops.preprocess import gpu_transform_batch
+from waternet_amd.ops import ext
+from waternet_amd.utils.metrics import (
+    peak_signal_noise_ratio,
+    structural_similarity_index_measure,
+)
+
+DEV = "cuda:0"
+STEPS = 150
+BS, H, W = 8, 64, 64
+
+
+def make_data():
+    rng = np.random.default_rng(0)
+    # fixed small synthetic set: smooth ramps + noise so SSIM is meaningful
+    yy, xx = np.mgrid[0:H, 0:W]
+    base = ((yy + xx) * 255 / (H + W)).astype(np.uint8)
+    raws, refs = [], []
+    for i in range(4):
+        noise = rng.integers(0, 80, size=(BS, H, W, 3), dtype=np.uint8)
+        raw = np.clip(base[None, :, :, None] * 0.5 + noise, 0, 255)
+        ref = np.clip(base[None, :, :, None] * 1.0 + noise * 0.3, 0, 255)
+        raws.append(raw.astype(np.uint8))
+        refs.append(ref.astype(np.uint8))
+    return raws, refs
+
+
+def run(eager: bool):
+    os.environ["WATERNET_AMD_EAGER"] = "1" if eager else "0"
+    torch.manual_seed(0)
+    model = WaterNet().to(DEV)
+    vgg = PerceptualModel(seed=1234).to(DEV).eval()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    raws, refs = make_data()
+    e = ext()
+    curve = []
+    for step in range(STEPS):
+        raw = torch.from_numpy(raws[step % 4]).to(DEV)
+        ref = torch.from_numpy(refs[step % 4]).to(DEV)
+        wb, gc, he = gpu_transform_batch(raw)
+        raw_f, wb_f, gc_f, he_f = (e.u8_to_nchw(t) for t in (raw, wb, gc, he))
+        ref_f = e.u8_to_nchw(ref)
+        out = model(raw_f, wb_f, he_f, gc_f)
+        loss, ploss, mloss = composite_loss(out, ref_f, vgg)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        if step % 10 == 0 or step == STEPS - 1:
+            with torch.no_grad():
+                ssim = structural_similarity_index_measure(
+                    preds=out.detach(), target=ref_f).item()
+                psnr = peak_signal_noise_ratio(
+                    preds=out.detach(), target=ref_f, data_range=1.0).item()
+            curve.append({"step": step, "loss": loss.item(),
+                          "mse255": mloss.item(), "ssim": ssim,
+                          "psnr": psnr})
+    return curve
+
+
+native = run(eager=False)
+eager = run(eager=True)
+result = {"steps": STEPS, "native_bf16": native, "eager_fp32": eager}
+outp = pathlib.Path(__file__).resolve().parent.parent / "profiles" / "convergence.json"
+outp.write_text(json.dumps(result, indent=1))
+nf, ef = native[-1], eager[-1]
+print("final native:", nf)
+print("final eager :", ef)
+rel = abs(nf["loss"] - ef["loss"]) / (abs(ef["loss"]) + 1e-9)
+print(f"final loss rel diff: {rel:.3f}")
+assert nf["loss"] < native[0]["loss"] * 0.7, "native loss did not decrease"
+assert ef["loss"] < eager[0]["loss"] * 0.7, "eager loss did not decrease"
+assert rel < 0.15, "native and eager trajectories diverged"
+assert abs(nf["psnr"] - ef["psnr"]) < 1.5
+print("CONVERGENCE PARITY OK")
