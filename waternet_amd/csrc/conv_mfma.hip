@@ -1302,7 +1302,10 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     const int base = gx * gy;
     int gz = 1;
     if (base < 320 && nk >= 16) gz = std::min(std::max(1, 512 / base), nk / 4);
-    gz = std::min(gz, 8);  // finalize reads gz slabs; >8 is latency-negative
+    // finalize's unrolled reads tolerate more slabs for the tiniest
+    // (grid-limited) shapes; snap >8 to the instantiated 12/16 variants
+    gz = std::min(gz, base < 64 ? 16 : 8);
+    if (gz > 8) gz = (gz >= 16) ? 16 : 12;
     if (gz > 1) {
       auto y32 = at::empty({gz, M, (long)Kp}, x.options().dtype(at::kFloat));
       const size_t lds5 =
@@ -1328,6 +1331,8 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
         case 5: fin(std::integral_constant<int, 5>{}); break;
         case 6: fin(std::integral_constant<int, 6>{}); break;
         case 7: fin(std::integral_constant<int, 7>{}); break;
+        case 12: fin(std::integral_constant<int, 12>{}); break;
+        case 16: fin(std::integral_constant<int, 16>{}); break;
         default: fin(std::integral_constant<int, 8>{}); break;
       }
     } else {
