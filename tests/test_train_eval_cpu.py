@@ -66,17 +66,20 @@ def test_training_reduces_loss():
 
 def test_train_cli_synthetic(tmp_path):
     """Run the real train.py CLI for 1 epoch on synthetic data."""
+    import os
+
+    env = dict(os.environ, WATERNET_TRAINING_DIR=str(tmp_path / "training"))
     out = subprocess.run(
         [
             sys.executable, "train.py", "--epochs", "1", "--batch-size", "2",
             "--height", "32", "--width", "32", "--synthetic", "4",
             "--seed", "0",
         ],
-        cwd=REPO, capture_output=True, text=True, timeout=600,
+        cwd=REPO, capture_output=True, text=True, timeout=600, env=env,
     )
     assert out.returncode == 0, out.stderr
     # Find the most recent training dir and check artifacts
-    training = REPO / "training"
+    training = tmp_path / "training"
     runs = sorted(
         (p for p in training.iterdir() if p.stem.isdecimal()),
         key=lambda p: int(p.stem),

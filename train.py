@@ -72,7 +72,10 @@ def main(argv=None):
     rank0 = dist_env.rank == 0
 
     projectroot = Path(__file__).parent
-    outputdir = projectroot / "training"
+    # savedir root: training/ beside train.py (reference behavior), or
+    # $WATERNET_TRAINING_DIR (tests / packaged installs)
+    outputdir = Path(os.environ.get("WATERNET_TRAINING_DIR",
+                                    projectroot / "training"))
     torch.manual_seed(0)  # always, as the reference (train.py:160)
     if args.seed is not None:
         torch.manual_seed(args.seed)
