@@ -22,7 +22,7 @@ from waternet_amd.ops.functional import (
 
 
 class WaterNetNativeState:
-    """Per-model ConvSpec table (packed-weight caches) + branch streams."""
+    """Per-model ConvSpec table (packed-weight caches)."""
 
     def __init__(self, model):
         cmg = model.cmg
@@ -37,14 +37,6 @@ class WaterNetNativeState:
                 ConvSpec(getattr(ref, f"conv{i + 1}"), ACT_RELU)
                 for i in range(3)
             ]
-        # One stream per refiner branch: the branches are independent of
-        # the CMG chain and of each other, and their Kp=32 convs underfill
-        # the 256 CUs — forking them overlaps both the forwards AND the
-        # backwards (autograd replays each op on its forward stream).
-        self.branch_streams = (
-            [torch.cuda.Stream() for _ in range(3)]
-            if torch.cuda.is_available() else None
-        )
 
     def mark_dirty(self):
         for s in self.cmg_specs:
@@ -74,35 +66,18 @@ def waternet_forward_native(model, x, wb, ce, gc):
                                 gc.float().contiguous())
     cmg_in, rwb_in, rce_in, rgc_in = inputs
 
-    branch_inputs = (("wb_refiner", rwb_in), ("ce_refiner", rce_in),
-                     ("gc_refiner", rgc_in))
+    t = cmg_in
+    for spec in st.cmg_specs:
+        t = conv_bias_act(t, spec)
+    maps = t
+
     refined = []
-    if st.branch_streams is not None and x.is_cuda:
-        main = torch.cuda.current_stream()
-        # fork: each refiner branch on its own stream, CMG on main
-        for s, (name, rin) in zip(st.branch_streams, branch_inputs):
-            s.wait_stream(main)
-            with torch.cuda.stream(s):
-                r = rin
-                for spec in st.refiner_specs[name]:
-                    r = conv_bias_act(r, spec)
-            refined.append(r)
-        t = cmg_in
-        for spec in st.cmg_specs:
-            t = conv_bias_act(t, spec)
-        maps = t
-        for s in st.branch_streams:
-            main.wait_stream(s)
-    else:
-        t = cmg_in
-        for spec in st.cmg_specs:
-            t = conv_bias_act(t, spec)
-        maps = t
-        for name, rin in branch_inputs:
-            r = rin
-            for spec in st.refiner_specs[name]:
-                r = conv_bias_act(r, spec)
-            refined.append(r)
+    for name, rin in (("wb_refiner", rwb_in), ("ce_refiner", rce_in),
+                      ("gc_refiner", rgc_in)):
+        r = rin
+        for spec in st.refiner_specs[name]:
+            r = conv_bias_act(r, spec)
+        refined.append(r)
 
     out_nhwc = GatedFusion.apply(maps, refined[0], refined[1], refined[2])
     return NhwcToNchw.apply(out_nhwc, 3)
