@@ -65,8 +65,9 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const long M = (long)N * H * W;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // SPLITK uses a 5-deep buffer ring (see below), the 2-phase path 2.
-  constexpr int NBUF = SPLITK ? 5 : 2;
+  // SPLITK uses a 5-deep buffer ring (see below); the 2-phase path uses
+  // 2 stage buffers of DK 32-K steps each (DK=2 for the narrow tiles).
+  constexpr int NBUF = SPLITK ? 5 : ((BN <= 32) ? 4 : 2);
   bf16_t* lA = reinterpret_cast<bf16_t*>(smem);              // NBUF x BM*32
   bf16_t* lB = lA + NBUF * BM * 32;                          // NBUF x LB
 
@@ -219,32 +220,46 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   } else {
     // 2-phase glds pipeline (guide §5 "minimum 2-phase"): issue next
     // tile's DMA first, ds_read + MFMA the current buffer, one
-    // vmcnt(0)+barrier per K-step (the __syncthreads drains the DMA).
-    stage(0, ks0 * 32);
+    // vmcnt(0)+barrier per staged tile (the __syncthreads drains the DMA).
+    // Narrow tiles (BN<=32) do only FM*FN<=4 MFMAs per 32-K step, so they
+    // stage DK=2 steps per barrier to amortize it (K tails are zero-filled
+    // by the staging, so no edge code).
+    constexpr int DK = (BN <= 32) ? 2 : 1;
+#pragma unroll
+    for (int h = 0; h < DK; ++h) stage(h, (ks0 + h) * 32);
     __syncthreads();
     int cur = 0;
-    for (int ks = ks0; ks < nk; ++ks) {
-      if (ks + 1 < nk) stage(cur ^ 1, (ks + 1) * 32);
-      bf16x8 aF[FM], bF[FN];
+    const int nk2 = (nk - ks0 + DK - 1) / DK;
+    for (int ks2 = 0; ks2 < nk2; ++ks2) {
+      if (ks2 + 1 < nk2) {
 #pragma unroll
-      for (int fm = 0; fm < FM; ++fm) {
-        int mfG = wr * FM + fm;
-        aF[fm] = *reinterpret_cast<const bf16x8*>(
-            lA + cur * (BM * 32) + ((mfG * 4 + lg) * 16 + li) * 8);
+        for (int h = 0; h < DK; ++h)
+          stage((cur ^ 1) * DK + h, (ks0 + (ks2 + 1) * DK + h) * 32);
       }
 #pragma unroll
-      for (int fn = 0; fn < FN; ++fn) {
-        int nfG = wc * FN + fn;
-        bF[fn] = *reinterpret_cast<const bf16x8*>(
-            lB + cur * LB + ((nfG * 4 + lg) * 16 + li) * 8);
+      for (int h = 0; h < DK; ++h) {
+        bf16x8 aF[FM], bF[FN];
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm) {
+          int mfG = wr * FM + fm;
+          aF[fm] = *reinterpret_cast<const bf16x8*>(
+              lA + (cur * DK + h) * (BM * 32) +
+              ((mfG * 4 + lg) * 16 + li) * 8);
+        }
+#pragma unroll
+        for (int fn = 0; fn < FN; ++fn) {
+          int nfG = wc * FN + fn;
+          bF[fn] = *reinterpret_cast<const bf16x8*>(
+              lB + (cur * DK + h) * LB + ((nfG * 4 + lg) * 16 + li) * 8);
+        }
+#pragma unroll
+        for (int fm = 0; fm < FM; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < FN; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
       }
-#pragma unroll
-      for (int fm = 0; fm < FM; ++fm)
-#pragma unroll
-        for (int fn = 0; fn < FN; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aF[fm], bF[fn], acc[fm][fn], 0, 0, 0);
-      if (ks + 1 < nk) __syncthreads();
+      if (ks2 + 1 < nk2) __syncthreads();
       cur ^= 1;
     }
   }
@@ -1295,7 +1310,9 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     constexpr int BN = decltype(bn_const)::value;
     const int gy = (Kp + BN - 1) / BN;
     constexpr int NBS = (BN * 4 + 255) / 256;
-    const size_t lds = (2 * 128 * 32 + 2 * NBS * 256 * 8) * sizeof(bf16_t);
+    constexpr int DKH = (BN <= 32) ? 2 : 1;
+    const size_t lds =
+        (2 * DKH * 128 * 32 + 2 * DKH * NBS * 256 * 8) * sizeof(bf16_t);
     // Small-M shapes (e.g. VGG 14^2/7^2 layers at bs=16) leave most of the
     // 256 CUs idle; split the K loop across gz slices into fp32 partials,
     // then finalize bias+act+bf16 in a second tiny pass.
