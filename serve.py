@@ -1,0 +1,102 @@
+"""Minimal production serving endpoint for the WaterNet engine.
+
+    python serve.py --host 0.0.0.0 --port 8000 --weights training/0/last.pt
+
+POST /enhance with a PNG/JPEG body (or multipart file) returns the enhanced
+image as PNG. GET /healthz reports device + model status. On a ROCm GPU the
+per-request pipeline is the hipGraph-captured InferenceEngine (fixed-shape
+graphs cached per resolution); on CPU it falls back to the eager path.
+"""
+
+import argparse
+import io
+from typing import Optional
+
+import numpy as np
+import torch
+
+from waternet_amd.models.waternet import WaterNet
+
+
+class Server:
+    def __init__(self, weights: Optional[str] = None, device: Optional[str] = None):
+        self.device = torch.device(
+            device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+        )
+        torch.manual_seed(0)
+        self.model = WaterNet()
+        if weights:
+            self.model.load_state_dict(torch.load(weights, map_location="cpu"))
+        self.model.to(self.device).eval()
+        self._engines = {}  # (H, W) -> InferenceEngine (GPU only)
+
+    def enhance(self, rgb_u8: np.ndarray) -> np.ndarray:
+        h, w = rgb_u8.shape[:2]
+        if self.device.type == "cuda" and h % 8 == 0 and w % 8 == 0:
+            eng = self._engines.get((h, w))
+            if eng is None:
+                from waternet_amd.engine.inferencer import InferenceEngine
+
+                eng = InferenceEngine(self.model, h, w, device=self.device)
+                self._engines[(h, w)] = eng
+            return eng.infer_frame(rgb_u8)
+        # CPU / odd-size fallback: reference transforms + eager forward
+        from waternet_amd.data.bridge import arr2ten, ten2arr
+        from waternet_amd.data.transforms import transform
+
+        wb, gc, he = transform(rgb_u8)
+        with torch.no_grad():
+            out = self.model(
+                arr2ten(rgb_u8, True).to(self.device),
+                arr2ten(wb, True).to(self.device),
+                arr2ten(he, True).to(self.device),
+                arr2ten(gc, True).to(self.device),
+            )
+        return ten2arr(out)[0]
+
+
+def create_app(weights: Optional[str] = None, device: Optional[str] = None):
+    from fastapi import FastAPI, Request, Response
+
+    from PIL import Image
+
+    app = FastAPI(title="waternet-mi355x")
+    server = Server(weights=weights, device=device)
+
+    @app.get("/healthz")
+    def healthz():
+        return {
+            "status": "ok",
+            "device": str(server.device),
+            "engines": list(map(list, server._engines.keys())),
+        }
+
+    @app.post("/enhance")
+    async def enhance(request: Request):
+        body = await request.body()
+        img = Image.open(io.BytesIO(body)).convert("RGB")
+        rgb = np.asarray(img)
+        out = server.enhance(rgb)
+        buf = io.BytesIO()
+        Image.fromarray(out).save(buf, format="PNG")
+        return Response(content=buf.getvalue(), media_type="image/png")
+
+    return app
+
+
+def main(argv=None):
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--host", default="127.0.0.1")
+    parser.add_argument("--port", type=int, default=8000)
+    parser.add_argument("--weights", default=None)
+    parser.add_argument("--device", default=None)
+    args = parser.parse_args(argv)
+
+    import uvicorn
+
+    uvicorn.run(create_app(args.weights, args.device), host=args.host,
+                port=args.port)
+
+
+if __name__ == "__main__":
+    main()
