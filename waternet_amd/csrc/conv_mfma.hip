@@ -1008,13 +1008,18 @@ __global__ __launch_bounds__(256, 4) void k_wgrad_smallk(
     unsigned oy = magic_div(rem, mulW);
     int ox = (int)(rem - oy * (unsigned)W);
     int iy = (int)oy + dy_ - PAD, ix = ox + dx_ - PAD;
-    if (iy < 0 || iy >= H || ix < 0 || ix >= W) continue;
+    // branch-free halo: clamped load + zero mask — a `continue` here makes
+    // hipcc branch around the loads and drain vmcnt per element (guide §5
+    // trap 4c), serializing the whole m loop on load latency
+    const bool valid = iy >= 0 && iy < H && ix >= 0 && ix < W;
+    const int iyc = min(max(iy, 0), H - 1), ixc = min(max(ix, 0), W - 1);
     const bf16x8 xv = *reinterpret_cast<const bf16x8*>(
-        X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + cg * 8);
+        X + (((long)((int)n * H + iyc) * W + ixc) << log2Cp) + cg * 8);
     const bf16_t* dyp = dY + m * Kp;
+    const float mask = valid ? 1.f : 0.f;
 #pragma unroll
     for (int k = 0; k < KMAX; ++k) {
-      const float d = bf2f(dyp[k]);
+      const float d = mask * bf2f(dyp[k]);
 #pragma unroll
       for (int e = 0; e < 8; ++e) acc[k][e] += d * bf2f(xv[e]);
     }
