@@ -33,3 +33,23 @@ def test_enhance_roundtrip(client):
     assert r.status_code == 200
     out = np.asarray(Image.open(io.BytesIO(r.content)).convert("RGB"))
     assert out.shape == img.shape and out.dtype == np.uint8
+
+
+@pytest.mark.gpu
+def test_serve_gpu_engine():
+    """On a GPU the server routes /8-divisible frames through the hipGraph
+    InferenceEngine and caches one engine per resolution."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    from serve import Server
+
+    s = Server()
+    rng = np.random.default_rng(1)
+    img = rng.integers(0, 256, size=(64, 80, 3), dtype=np.uint8)
+    out1 = s.enhance(img)
+    out2 = s.enhance(img)
+    assert out1.shape == img.shape
+    assert (out1 == out2).all()
+    assert (64, 80) in s._engines
