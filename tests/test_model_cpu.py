@@ -101,3 +101,24 @@ def test_backward_produces_grads():
     grads = [p.grad for p in model.parameters()]
     assert all(g is not None for g in grads)
     assert all(torch.isfinite(g).all() for g in grads)
+
+
+def test_torch_hub_local_load():
+    """torch.hub.load(source='local') end-to-end through hubconf.py."""
+    import pathlib
+
+    import torch
+
+    repo = str(pathlib.Path(__file__).resolve().parent.parent)
+    pre, post, model = torch.hub.load(
+        repo, "waternet", source="local", pretrained=False, device="cpu"
+    )
+    import numpy as np
+
+    rng = np.random.default_rng(0)
+    rgb = rng.integers(0, 256, size=(32, 32, 3), dtype=np.uint8)
+    rgb_t, wb_t, he_t, gc_t = pre(rgb)
+    with torch.no_grad():
+        out = model(rgb_t, wb_t, he_t, gc_t)
+    arr = post(out)
+    assert arr.shape == (1, 32, 32, 3) and arr.dtype == np.uint8
