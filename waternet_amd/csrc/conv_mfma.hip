@@ -836,51 +836,64 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
       loadB(next * CH);
     }
     if constexpr (M32) {
-      // two half-chunk batches (2 x 16-m ksteps each) keep the tr-read
-      // register footprint at 16 bf16x4
+      // fine-grained per-kstep interleave: tr-read kstep k+1 while the
+      // MFMAs of kstep k issue, with COUNTED lgkmcnt so waves drift apart
+      // instead of entering barrier-aligned MFMA bursts together (PMC:
+      // 61% SQ_WAIT_INST_ANY at only ~24% MFMA-pipe duty).
       constexpr int KST = CH / 16;  // 32x32x16 ksteps per chunk
+      constexpr int RPK = 2 * (FK32 + FR32);  // tr reads per kstep
+      bf16x4 aT[2][FK32][2], bT[2][FR32][2];  // 2-deep kstep ring
+      auto rdk = [&](int kst, int ring) {
 #pragma unroll
-      for (int hb = 0; hb < 2; ++hb) {
-        bf16x4 aT[KST / 2][FK32][2], bT[KST / 2][FR32][2];
-#pragma unroll
-        for (int s = 0; s < KST / 2; ++s) {
-          const int kst = hb * (KST / 2) + s;
-#pragma unroll
-          for (int f = 0; f < FK32; ++f) {
-            const unsigned base =
-                aTrM + (unsigned)((wr * (BK / WR / 16) + f * 2) * KBS +
-                                  kst * 4 * TR_MBS) * 2;
-            aT[s][f][0] = ds_tr16(base);
-            aT[s][f][1] = ds_tr16(base + TR_MBS * 2);
-          }
-#pragma unroll
-          for (int f = 0; f < FR32; ++f) {
-            const unsigned base =
-                bTrM + (unsigned)((wc * (BR / WC / 16) + f * 2) * KBS +
-                                  kst * 4 * TR_MBS) * 2;
-            bT[s][f][0] = ds_tr16(base);
-            bT[s][f][1] = ds_tr16(base + TR_MBS * 2);
-          }
+        for (int f = 0; f < FK32; ++f) {
+          const unsigned base =
+              aTrM + (unsigned)((wr * (BK / WR / 16) + f * 2) * KBS +
+                                kst * 4 * TR_MBS) * 2;
+          aT[ring][f][0] = ds_tr16(base);
+          aT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+        for (int f = 0; f < FR32; ++f) {
+          const unsigned base =
+              bTrM + (unsigned)((wc * (BR / WC / 16) + f * 2) * KBS +
+                                kst * 4 * TR_MBS) * 2;
+          bT[ring][f][0] = ds_tr16(base);
+          bT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
+        }
+      };
+      rdk(0, 0);
+#pragma unroll
+      for (int kst = 0; kst < KST; ++kst) {
+        const int ring = kst & 1;
+        if (kst + 1 < KST) rdk(kst + 1, ring ^ 1);
+        // wait for THIS kstep's reads only; the next kstep's RPK stay
+        // in flight under the MFMAs
+        if (kst + 1 < KST) {
+          if constexpr (RPK == 8)
+            asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+          else if constexpr (RPK == 6)
+            asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
+          else
+            asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        }
         __builtin_amdgcn_sched_barrier(0);
-        if (hb == 1) {
-          __syncthreads();  // ALL reads done; LDS free for restaging
+        if (kst == KST - 1) {
+          __syncthreads();  // ALL tr reads done; LDS free for restaging
           if (next < nChunks) writeTiles();
         }
 #pragma unroll
-        for (int s = 0; s < KST / 2; ++s)
+        for (int fa = 0; fa < FK32; ++fa)
 #pragma unroll
-          for (int fa = 0; fa < FK32; ++fa)
-#pragma unroll
-            for (int fb = 0; fb < FR32; ++fb) {
-              const bf16x8 av = __builtin_shufflevector(
-                  aT[s][fa][0], aT[s][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
-              const bf16x8 bv = __builtin_shufflevector(
-                  bT[s][fb][0], bT[s][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
-              acc32[fa][fb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                  av, bv, acc32[fa][fb], 0, 0, 0);
-            }
+          for (int fb = 0; fb < FR32; ++fb) {
+            const bf16x8 av = __builtin_shufflevector(
+                aT[ring][fa][0], aT[ring][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
+            const bf16x8 bv = __builtin_shufflevector(
+                bT[ring][fb][0], bT[ring][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
+            acc32[fa][fb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                av, bv, acc32[fa][fb], 0, 0, 0);
+          }
       }
     } else {
       // tr-read ALL fragments for this chunk into registers (guide T10)
