@@ -679,7 +679,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
     const bf16_t* __restrict__ X,   // (N,H,W,Cp)
     float* __restrict__ dW,         // (K, C, KS, KS) fp32, pre-zeroed
     int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C,
-    int splitm, unsigned long long mulHW, unsigned long long mulW) {
+    int splitm, unsigned long long mulHW, unsigned long long mulW,
+    const bf16_t* __restrict__ Zero16w) {
   static_assert(WN_MFMA_KMAP == 0, "wgrad staging assumes KMAP 0");
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;
@@ -691,16 +692,28 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   constexpr int WC = 4 / WR;               // wave cols (rsc dim)
   constexpr int FK = BK / WR / 16;  // k fragments per wave
   constexpr int FR = BR / WC / 16;  // rsc fragments per wave
-  constexpr int ATOT = CH * BK / 8;         // total A vec8 slots (>=128)
-  constexpr int ASLOT = (ATOT + 255) / 256; // A vec8 loads per thread
-  constexpr int BSLOT = CH * BR / 8 / 256;  // B vec8 loads per thread
   const int KG = RS * Cp;
   const long M = (long)N * H * W;
   const int HW = H * W;
 
+  // glds staging: the tr image's linear 8-element chunk index sigma maps
+  // IDENTICALLY to tr_addr (sigma*8 == tr_addr(m, col) for m = mblk*4 +
+  // (r>>1), col = kblk*16 + (r&1)*8, sigma = kblk*CPK + mblk*9 + r), so
+  // global_load_lds writes the image lane-linearly while each lane's
+  // SOURCE gathers the right 16 B of dY/X (pad/halo lanes read Zero16).
+  // Double-buffered: the next chunk's DMA issues before this chunk's
+  // MFMAs and drains at the single trailing barrier.
+  constexpr int CPK = KBS / 8;               // chunks per kblk
+  constexpr int AC = (BK / 16) * CPK;        // A image chunks
+  constexpr int BC = (BR / 16) * CPK;        // B image chunks
+  constexpr int SA = (AC + 255) / 256;
+  constexpr int SB = (BC + 255) / 256;
+  constexpr int ABUF = SA * 256 * 8;         // elems per buffer (padded so
+  constexpr int BBUF = SB * 256 * 8;         //  out-of-image slots land in-pad)
+
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // (BK/16) x KBS
-  bf16_t* lB = lA + (BK / 16) * KBS;              // (BR/16) x KBS
+  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);   // 2 x ABUF
+  bf16_t* lB = lA + 2 * ABUF;                     // 2 x BBUF
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -710,79 +723,58 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const int kt0 = blockIdx.x * BK;
   const int rt0 = blockIdx.y * BR;
 
-  // ---- static staging geometry ----
-  // A slot s: jg = slot % (BK/8) (k group), ml = slot / (BK/8) (m row)
-  int aJg[ASLOT], aMl[ASLOT];
-  bool aV[ASLOT];
+  // ---- static slot geometry (sigma -> image position -> source) ----
+  int aM_[SA], aCol[SA];
+  bool aPad[SA];
 #pragma unroll
-  for (int s = 0; s < ASLOT; ++s) {
-    int slot = tid + s * 256;
-    aV[s] = slot < ATOT;
-    slot = aV[s] ? slot : 0;
-    aJg[s] = slot % (BK / 8);
-    aMl[s] = slot / (BK / 8);
+  for (int s = 0; s < SA; ++s) {
+    const int sg = tid + s * 256;
+    const int kblk = sg / CPK, c = sg % CPK;
+    const int mblk = c / 9, r = c % 9;
+    aPad[s] = (sg >= AC) || (r == 8) || (c >= (CH / 4) * 9);
+    aM_[s] = mblk * 4 + (r >> 1);
+    aCol[s] = kt0 + kblk * 16 + (r & 1) * 8;
   }
-  // B slot s: rsc group + m row; decode tap/c once (fixed per thread)
-  int bMl[BSLOT], bC[BSLOT], bDy[BSLOT], bDx[BSLOT];
-  bool bIn[BSLOT];
+  int bM_[SB], bC_[SB], bDy[SB], bDx[SB];
+  bool bPad[SB];
 #pragma unroll
-  for (int s = 0; s < BSLOT; ++s) {
-    int slot = tid + s * 256;
-    int jg = slot % (BR / 8);
-    bMl[s] = slot / (BR / 8);
-    int rsc = rt0 + jg * 8;
-    bIn[s] = rsc < KG;
-    int tap = rsc >> log2Cp;
-    bC[s] = rsc & (Cp - 1);
+  for (int s = 0; s < SB; ++s) {
+    const int sg = tid + s * 256;
+    const int kblk = sg / CPK, c = sg % CPK;
+    const int mblk = c / 9, r = c % 9;
+    const int rsc = rt0 + kblk * 16 + (r & 1) * 8;
+    bPad[s] = (sg >= BC) || (r == 8) || (c >= (CH / 4) * 9) || (rsc >= KG);
+    bM_[s] = mblk * 4 + (r >> 1);
+    const int tap = rsc >> log2Cp;
+    bC_[s] = rsc & (Cp - 1);
     bDy[s] = tap / KS;
     bDx[s] = tap - bDy[s] * KS;
   }
 
-  bf16x8 aR[ASLOT], bR[BSLOT];
-
-  auto loadA = [&](long m0) {
+  auto stage = [&](int buf, long mbase) {
 #pragma unroll
-    for (int s = 0; s < ASLOT; ++s) {
-      bf16x8 v = {};
-      long m = m0 + aMl[s];
-      if (m < M) {
-        int kg = kt0 + aJg[s] * 8;
-        if (kg + 7 < Kp)
-          v = *reinterpret_cast<const bf16x8*>(dY + m * Kp + kg);
-      }
-      aR[s] = v;
+    for (int s = 0; s < SA; ++s) {
+      const bf16_t* src = Zero16w;
+      const long m = mbase + aM_[s];
+      if (!aPad[s] && m < M) src = dY + m * Kp + aCol[s];
+      __builtin_amdgcn_global_load_lds(
+          src, lA + buf * ABUF + (tid + s * 256) * 8, 16, 0, 0);
     }
-  };
-  auto loadB = [&](long m0) {
 #pragma unroll
-    for (int s = 0; s < BSLOT; ++s) {
-      bf16x8 v = {};
-      long m = m0 + bMl[s];
-      if (m < M && bIn[s]) {
+    for (int s = 0; s < SB; ++s) {
+      const bf16_t* src = Zero16w;
+      const long m = mbase + bM_[s];
+      if (!bPad[s] && m < M) {
         unsigned n = magic_div((unsigned)m, mulHW);
         unsigned rem = (unsigned)m - n * (unsigned)HW;
         unsigned oy = magic_div(rem, mulW);
         int ox = (int)(rem - oy * (unsigned)W);
         int iy = (int)oy + bDy[s] - PAD, ix = ox + bDx[s] - PAD;
         if (iy >= 0 && iy < H && ix >= 0 && ix < W)
-          v = *reinterpret_cast<const bf16x8*>(
-              X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + bC[s]);
+          src = X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + bC_[s];
       }
-      bR[s] = v;
-    }
-  };
-  auto writeTiles = [&]() {
-#pragma unroll
-    for (int s = 0; s < ASLOT; ++s) {
-      if (!aV[s]) continue;
-      *reinterpret_cast<bf16x8*>(
-          lA + tr_addr<KBS>(aMl[s], aJg[s] * 8)) = aR[s];
-    }
-#pragma unroll
-    for (int s = 0; s < BSLOT; ++s) {
-      int slot = tid + s * 256;
-      int jg = slot % (BR / 8);
-      *reinterpret_cast<bf16x8*>(lB + tr_addr<KBS>(bMl[s], jg * 8)) = bR[s];
+      __builtin_amdgcn_global_load_lds(
+          src, lB + buf * BBUF + (tid + s * 256) * 8, 16, 0, 0);
     }
   };
 
@@ -823,18 +815,16 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const unsigned bTrM = (unsigned)(unsigned long long)lB + trM32;
 
   long chunk = blockIdx.z;
+  int buf = 0;
   if (chunk < nChunks) {
-    loadA(chunk * CH);
-    loadB(chunk * CH);
-    writeTiles();
-    __syncthreads();
+    stage(0, chunk * CH);
+    __syncthreads();  // drains the DMA (vmcnt 0) + barrier
   }
   for (; chunk < nChunks; chunk += splitm) {
     const long next = chunk + splitm;
-    if (next < nChunks) {
-      loadA(next * CH);  // global -> regs, overlaps this chunk's math
-      loadB(next * CH);
-    }
+    if (next < nChunks) stage(buf ^ 1, next * CH);  // DMA under the math
+    const unsigned aOff = (unsigned)(buf * ABUF) * 2;
+    const unsigned bOff = (unsigned)(buf * BBUF) * 2;
     if constexpr (M32) {
       // fine-grained per-kstep interleave: tr-read kstep k+1 while the
       // MFMAs of kstep k issue, with COUNTED lgkmcnt so waves drift apart
@@ -847,16 +837,16 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 #pragma unroll
         for (int f = 0; f < FK32; ++f) {
           const unsigned base =
-              aTrM + (unsigned)((wr * (BK / WR / 16) + f * 2) * KBS +
-                                kst * 4 * TR_MBS) * 2;
+              aTrM + aOff + (unsigned)((wr * (BK / WR / 16) + f * 2) * KBS +
+                                       kst * 4 * TR_MBS) * 2;
           aT[ring][f][0] = ds_tr16(base);
           aT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
         }
 #pragma unroll
         for (int f = 0; f < FR32; ++f) {
           const unsigned base =
-              bTrM + (unsigned)((wc * (BR / WC / 16) + f * 2) * KBS +
-                                kst * 4 * TR_MBS) * 2;
+              bTrM + bOff + (unsigned)((wc * (BR / WC / 16) + f * 2) * KBS +
+                                       kst * 4 * TR_MBS) * 2;
           bT[ring][f][0] = ds_tr16(base);
           bT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
         }
@@ -879,10 +869,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
           asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_sched_barrier(0);
-        if (kst == KST - 1) {
-          __syncthreads();  // ALL tr reads done; LDS free for restaging
-          if (next < nChunks) writeTiles();
-        }
+
 #pragma unroll
         for (int fa = 0; fa < FK32; ++fa)
 #pragma unroll
@@ -902,23 +889,23 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
       for (int s2 = 0; s2 < CH / 32; ++s2) {
 #pragma unroll
         for (int f = 0; f < FK; ++f) {
-          const unsigned base = aTr0 + (unsigned)((wr * FK + f) * KBS +
-                                                  s2 * 8 * TR_MBS) * 2;
+          const unsigned base = aTr0 + aOff +
+                                (unsigned)((wr * FK + f) * KBS +
+                                           s2 * 8 * TR_MBS) * 2;
           aT[s2][f][0] = ds_tr16(base);
           aT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
         }
 #pragma unroll
         for (int f = 0; f < FR; ++f) {
-          const unsigned base = bTr0 + (unsigned)((wc * FR + f) * KBS +
-                                                  s2 * 8 * TR_MBS) * 2;
+          const unsigned base = bTr0 + bOff +
+                                (unsigned)((wc * FR + f) * KBS +
+                                           s2 * 8 * TR_MBS) * 2;
           bT[s2][f][0] = ds_tr16(base);
           bT[s2][f][1] = ds_tr16(base + TR_MBS * 2);
         }
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_sched_barrier(0);  // keep MFMAs below the wait
-      __syncthreads();  // all waves done reading; LDS free for restaging
-      if (next < nChunks) writeTiles();  // issues alongside the MFMAs below
 #pragma unroll
       for (int s2 = 0; s2 < CH / 32; ++s2)
 #pragma unroll
@@ -933,7 +920,8 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
                 av, bv, acc[fa][fb], 0, 0, 0);
           }
     }
-    if (next < nChunks) __syncthreads();  // writes visible before next reads
+    if (next < nChunks) __syncthreads();  // drains DMA + joins waves
+    buf ^= 1;
   }
 
   if constexpr (M32) {
@@ -1493,18 +1481,20 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   }
   const int KG = (int)(ks * ks) * Cp;
   const int BK = std::min(Kp, 128);
-  // m rows per chunk: 64 for the wide tiles; 128 for BK<=32 (those do only
-  // 8 MFMAs per wave between barriers at CH=64 — amortize the chunk
-  // overhead; LDS still fits 3 blocks/CU).
-  const int WG_CH = (BK <= 32) ? 128 : 64;
+  constexpr int WG_CH = 64;  // m rows per chunk (glds double-buffer budget)
   const int gx = (Kp + BK - 1) / BK, gy = (KG + 127) / 128;
   // split so gx*gy*split fills 256 CUs x ~3 resident blocks
   int split = std::max(1, 640 / std::max(1, gx * gy));
   const long nChunks = ((long)N * H * W + WG_CH - 1) / WG_CH;
   split = (int)std::min<long>(split, nChunks);
   hipStream_t stream = at::cuda::getCurrentHIPStream();
-  const size_t lds =
-      (size_t)(BK / 16 + 128 / 16) * (WG_CH / 4 * 72 + 16) * sizeof(bf16_t);
+  const int cpk = (WG_CH / 4 * 72 + 16) / 8;  // image chunks per kblk
+  const int sa = ((BK / 16) * cpk + 255) / 256;
+  const int sb = (8 * cpk + 255) / 256;
+  const size_t lds = (size_t)2 * (sa + sb) * 256 * 8 * sizeof(bf16_t);
+  static thread_local at::Tensor zero16w;
+  if (!zero16w.defined() || zero16w.device() != x.device())
+    zero16w = at::zeros({8}, x.options());
   static const bool use_m32 = [] {
     const char* e = getenv("WN_WGRAD_M32");
     return e == nullptr || atoi(e) != 0;  // default ON; 0 = 16x16 path
@@ -1512,7 +1502,7 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   auto launch = [&](auto ks_const, auto bk_const) {
     constexpr int KSV = decltype(ks_const)::value;
     constexpr int BKV = decltype(bk_const)::value;
-    constexpr int CHV = (BKV <= 32) ? 128 : 64;
+    constexpr int CHV = 64;
     auto go = [&](auto m32_const) {
       constexpr bool M32V = decltype(m32_const)::value;
       hipLaunchKernelGGL((k_conv_wgrad<KSV, BKV, CHV, M32V>),
@@ -1521,7 +1511,8 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
                          (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
                          N, H, W, Cp, log2i(Cp), Kp, K, C, split,
                          MagicDiv::make((unsigned)(H * W)).mul,
-                         MagicDiv::make((unsigned)W).mul);
+                         MagicDiv::make((unsigned)W).mul,
+                         (const bf16_t*)zero16w.data_ptr());
     };
     if (BKV >= 32 && use_m32)
       go(std::integral_constant<bool, (BKV >= 32)>{});
