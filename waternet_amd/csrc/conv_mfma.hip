@@ -622,13 +622,19 @@ __global__ void k_splitk_finalize(const float* __restrict__ Y32,
 // 8-lane service group hit 8 distinct bank quads).
 //
 //   - BK x 128 output tile per block (BK = min(Kp,128)), 2x2 / 1x4 waves
-//   - m-chunks of 64 (two MFMA k-steps); single LDS buffer: tr-read all
-//     fragments into registers, barrier, restage next chunk's tiles
-//     back-to-back with this chunk's MFMAs, barrier.
-//   - split-m z dimension sized to fill 256 CUs x 2 blocks; gy (rsc tiles)
-//     varies fastest so co-resident blocks share dY/X chunks through L2.
+//   - m-chunks of 64; DOUBLE-BUFFERED global_load_lds staging: the linear
+//     chunk index of the tr image maps identically to tr_addr, so the DMA
+//     writes the image lane-linearly while per-lane sources gather the
+//     right 16 B of dY / halo-masked X (pad lanes read a zero buffer).
+//     The next chunk's DMA issues before this chunk's MFMAs and drains at
+//     the single trailing barrier.
+//   - default MFMA is v_mfma_f32_32x32x16_bf16 (M32) with a per-kstep
+//     counted-lgkmcnt tr-read interleave; WN_WGRAD_M32=0 selects the
+//     16x16x32 path.
+//   - split-m z dimension sized to fill the CUs; gy (rsc tiles) varies
+//     fastest so co-resident blocks share dY/X chunks through L2.
 //   - fp32 atomicAdd epilogue into the NCHW fp32 grad tensor (pre-zeroed).
-// Requires the KMAP-0 fragment layout (frag_k16(g,e) = g*8+e).
+// Requires the KMAP-0 fragment layout (lane group g consumes k = g*8+e).
 // ---------------------------------------------------------------------------
 
 constexpr int TR_MBS = 72;    // mblk stride (36 dwords; 2 mblks = 8 mod 64
