@@ -66,11 +66,9 @@ class BenchTrainer:
         # on-device metric accumulators
         self.metric_sums = torch.zeros(5, dtype=torch.float64,
                                        device=self.device)
-        self.n_steps_accum = 0
         self._i = 0
         self._graph = None
         self._use_graph = use_graph
-        self._last_lr = lr
         self._comm_stream = (torch.cuda.Stream()
                              if self.device.type == "cuda" else None)
         self._vgg_stream = (torch.cuda.Stream()
