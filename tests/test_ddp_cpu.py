@@ -53,9 +53,11 @@ def _run_allreduce_worker(rank, world_size, port, q):
         q.put((rank, "ERROR", repr(e), None))
 
 
-def test_flat_bucket_reducer_world2():
-    world_size = 2
-    port = 29811
+import pytest
+
+
+@pytest.mark.parametrize("world_size,port", [(2, 29811), (4, 29815)])
+def test_flat_bucket_reducer(world_size, port):
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [
@@ -72,7 +74,8 @@ def test_flat_bucket_reducer_world2():
     psums = {r[1] for r in results}
     assert len({round(s, 6) for s in psums}) == 1, "broadcast_params failed"
     assert all(r[2] for r in results), "grads differ across ranks"
-    assert all(abs(r[3] - 0.5) < 1e-9 for r in results), "metric avg wrong"
+    expect = sum(range(world_size)) / world_size
+    assert all(abs(r[3] - expect) < 1e-9 for r in results), "metric avg wrong"
 
 
 def _run_shard_worker(rank, world_size, port, q):
