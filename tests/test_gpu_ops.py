@@ -115,7 +115,7 @@ def test_tr_read_lane_mapping(ext):
 
 
 @pytest.mark.parametrize("ks,C,K", [(7, 12, 128), (5, 128, 128), (3, 32, 3),
-                                    (1, 128, 64)])
+                                    (1, 128, 64), (5, 6, 32), (3, 64, 3)])
 def test_conv_backward_parity(ext, ks, C, K):
     """Full autograd through ConvBiasAct vs F.conv2d fp32 on bf16-rounded
     inputs: dx, dw, db."""
