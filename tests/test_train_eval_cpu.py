@@ -159,3 +159,31 @@ def test_train_resume_full_state(tmp_path, monkeypatch):
     lines = (savedir2 / "metrics-train.csv").read_text().strip().splitlines()
     assert len(lines) == 2  # header + 1 epoch
     json.loads((savedir2 / "config.json").read_text())
+
+
+def test_eval_reference_bug_mode(monkeypatch):
+    """WATERNET_REFERENCE_EVAL_BUG=1 replicates the reference's last-batch
+    perceptual (train.py:71,74); default mode reports the true mean."""
+    import torch
+
+    from waternet_amd.engine.trainer import eval_one_epoch
+    from waternet_amd.models.vgg import PerceptualModel
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(0)
+    model = WaterNet()
+    vgg = PerceptualModel()
+    batches = []
+    for i in range(3):
+        g = torch.Generator().manual_seed(i)
+        t = torch.rand(2, 3, 32, 32, generator=g)
+        r = torch.rand(2, 3, 32, 32, generator=g)
+        batches.append({"raw": t, "wb": t, "he": t, "gc": t, "ref": r})
+
+    fixed = eval_one_epoch(model, batches, "cpu", vgg)
+    monkeypatch.setenv("WATERNET_REFERENCE_EVAL_BUG", "1")
+    buggy = eval_one_epoch(model, batches, "cpu", vgg)
+    # buggy = last-batch/3; fixed = mean over 3 — they must differ and the
+    # other metrics must be identical
+    assert abs(fixed["perceptual_loss"] - buggy["perceptual_loss"]) > 1e-9
+    assert abs(fixed["mse"] - buggy["mse"]) < 1e-9

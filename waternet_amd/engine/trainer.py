@@ -97,6 +97,14 @@ def train_one_epoch(
 
 @torch.no_grad()
 def eval_one_epoch(model, val_dataloader, device, vgg_model):
+    import os
+
+    # Reference mode (WATERNET_REFERENCE_EVAL_BUG=1): replicate the
+    # reference's `=` instead of `+=` at train.py:71, which makes the
+    # reported val perceptual = last-batch value / n_batches — needed only
+    # when parity-scoring against the reference's own printed numbers
+    # (SURVEY §7 behavioral quirks).
+    replicate_bug = os.environ.get("WATERNET_REFERENCE_EVAL_BUG", "0") == "1"
     model.eval()
     epoch_metrics = {k: 0.0 for k in VAL_METRICS_NAMES}
     n_batches = len(val_dataloader)
@@ -106,7 +114,10 @@ def eval_one_epoch(model, val_dataloader, device, vgg_model):
         out = model(batch["raw"], batch["wb"], batch["he"], batch["gc"])
 
         ploss = perceptual_loss(out, batch["ref"], vgg_model)
-        epoch_metrics["perceptual_loss"] += ploss.item()
+        if replicate_bug:
+            epoch_metrics["perceptual_loss"] = ploss.item()
+        else:
+            epoch_metrics["perceptual_loss"] += ploss.item()
         epoch_metrics["mse"] += mse_loss(out, batch["ref"]).item()
         epoch_metrics["ssim"] += structural_similarity_index_measure(
             preds=out, target=batch["ref"]
