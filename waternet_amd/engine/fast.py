@@ -44,17 +44,29 @@ class BenchTrainer:
                                                      gamma=0.1)
         # synthetic uint8 data pool (deterministic per rank), kept in PINNED
         # HOST memory so every step pays the real host->HBM transfer the
-        # reference's dataloader pays
+        # reference's dataloader pays. The images are
+        # spatially-correlated noise (low-res noise bilinearly upsampled +
+        # fine noise), matching natural-image statistics rather than
+        # full-range white noise — the reference trains on photographs.
         rng = np.random.default_rng(seed)
         pin = self.device.type == "cuda"
+
+        def synth_images():
+            base = rng.integers(
+                0, 256,
+                size=(batch_size, max(height // 8, 1), max(width // 8, 1), 3),
+            ).astype(np.float32)
+            t = torch.from_numpy(base).permute(0, 3, 1, 2)
+            up = torch.nn.functional.interpolate(
+                t, size=(height, width), mode="bilinear", align_corners=False
+            ).permute(0, 2, 3, 1).numpy()
+            fine = rng.normal(0.0, 12.0, size=up.shape)
+            return torch.from_numpy(
+                np.clip(up + fine, 0, 255).astype(np.uint8))
+
         self.pool = []
         for _ in range(pool_size):
-            raw = torch.from_numpy(rng.integers(
-                0, 256, size=(batch_size, height, width, 3), dtype=np.uint8
-            ))
-            ref = torch.from_numpy(rng.integers(
-                0, 256, size=(batch_size, height, width, 3), dtype=np.uint8
-            ))
+            raw, ref = synth_images(), synth_images()
             if pin:
                 raw, ref = raw.pin_memory(), ref.pin_memory()
             self.pool.append((raw, ref))
