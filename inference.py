@@ -54,16 +54,35 @@ def make_savedir(name):
 
 
 def load_model(weights, device):
+    """--weights path, or the reference's auto-download scheme (reference
+    inference.py:15-21,103-109): fetch the published hash-named checkpoint
+    next to this script with torch.hub's check_hash; falls back to a
+    random-init warning when offline."""
     model = WaterNet()
     if weights is not None:
         with open(weights, "rb") as f:
             model.load_state_dict(torch.load(f, map_location="cpu"))
     else:
-        print(
-            "WARNING: no --weights given and no network for auto-download; "
-            "using random-init weights.",
-            file=sys.stderr,
-        )
+        import hubconf
+
+        wd = Path(__file__).resolve().parent
+        local = wd / hubconf.WEIGHTS_FILE
+        try:
+            if local.exists():
+                model.load_state_dict(torch.load(local, map_location="cpu"))
+            else:
+                sd = torch.hub.load_state_dict_from_url(
+                    hubconf.WEIGHTS_URL, map_location="cpu",
+                    model_dir=str(wd), file_name=hubconf.WEIGHTS_FILE,
+                    check_hash=True,
+                )
+                model.load_state_dict(sd)
+        except Exception as e:  # noqa: BLE001 — offline environment
+            print(
+                "WARNING: no --weights given and auto-download failed "
+                f"({e!r}); using random-init weights.",
+                file=sys.stderr,
+            )
     model.to(device).eval()
     return model
 

@@ -10,12 +10,20 @@ graphs cached per resolution); on CPU it falls back to the eager path.
 
 import argparse
 import io
+from collections import OrderedDict
 from typing import Optional
 
 import numpy as np
 import torch
 
 from waternet_amd.models.waternet import WaterNet
+
+# A graph-captured engine holds static buffers + activation workspace per
+# resolution; bound the cache so clients posting many distinct resolutions
+# cannot exhaust HBM (oldest engine evicted, its graph memory freed).
+MAX_ENGINES = 8
+MAX_DIM = 4096  # reject images beyond 4096 px per side (~50 MB fp32 @4k)
+MAX_BODY_BYTES = 64 * 1024 * 1024
 
 
 class Server:
@@ -28,17 +36,24 @@ class Server:
         if weights:
             self.model.load_state_dict(torch.load(weights, map_location="cpu"))
         self.model.to(self.device).eval()
-        self._engines = {}  # (H, W) -> InferenceEngine (GPU only)
+        self._engines = OrderedDict()  # (H, W) -> InferenceEngine LRU
 
     def enhance(self, rgb_u8: np.ndarray) -> np.ndarray:
         h, w = rgb_u8.shape[:2]
+        if h > MAX_DIM or w > MAX_DIM:
+            raise ValueError(
+                f"image {w}x{h} exceeds the {MAX_DIM}px per-side limit")
         if self.device.type == "cuda" and h % 8 == 0 and w % 8 == 0:
             eng = self._engines.get((h, w))
             if eng is None:
                 from waternet_amd.engine.inferencer import InferenceEngine
 
                 eng = InferenceEngine(self.model, h, w, device=self.device)
+                while len(self._engines) >= MAX_ENGINES:
+                    self._engines.popitem(last=False)
                 self._engines[(h, w)] = eng
+            else:
+                self._engines.move_to_end((h, w))
             return eng.infer_frame(rgb_u8)
         # CPU / odd-size fallback: reference transforms + eager forward
         from waternet_amd.data.bridge import arr2ten, ten2arr
@@ -74,9 +89,14 @@ def create_app(weights: Optional[str] = None, device: Optional[str] = None):
     @app.post("/enhance")
     async def enhance(request: Request):
         body = await request.body()
+        if len(body) > MAX_BODY_BYTES:
+            return Response(status_code=413, content="image body too large")
         img = Image.open(io.BytesIO(body)).convert("RGB")
         rgb = np.asarray(img)
-        out = server.enhance(rgb)
+        try:
+            out = server.enhance(rgb)
+        except ValueError as e:
+            return Response(status_code=413, content=str(e))
         buf = io.BytesIO()
         Image.fromarray(out).save(buf, format="PNG")
         return Response(content=buf.getvalue(), media_type="image/png")

@@ -91,3 +91,58 @@ def test_hubconf_tuple():
     arr = postprocess(out)
     assert arr.shape == (1, 32, 32, 3)
     assert arr.dtype == np.uint8
+
+
+def test_synthetic_dataset_raw_mode():
+    import torch
+
+    from waternet_amd.data.dataset import SyntheticUIEBDataset
+
+    ds = SyntheticUIEBDataset(n_images=3, im_height=16, im_width=16,
+                              raw_mode=True)
+    item = ds[0]
+    assert set(item.keys()) == {"raw", "ref"}
+    assert item["raw"].dtype == torch.uint8
+    assert item["raw"].shape == (16, 16, 3)
+    # collates to (B,H,W,3) uint8 for the fast engine's static buffers
+    loader = torch.utils.data.DataLoader(ds, batch_size=2)
+    batch = next(iter(loader))
+    assert batch["raw"].shape == (2, 16, 16, 3)
+    assert batch["raw"].dtype == torch.uint8
+
+
+def test_uieb_dataset_raw_mode(tmp_path):
+    import numpy as np
+    import torch
+    from PIL import Image
+
+    from waternet_amd.data.dataset import UIEBDataset
+
+    (tmp_path / "raw").mkdir()
+    (tmp_path / "ref").mkdir()
+    rng = np.random.default_rng(0)
+    for i in range(2):
+        for d in ("raw", "ref"):
+            Image.fromarray(rng.integers(
+                0, 256, size=(40, 40, 3), dtype=np.uint8
+            )).save(tmp_path / d / f"{i}.png")
+    ds = UIEBDataset(tmp_path / "raw", tmp_path / "ref", im_height=32,
+                     im_width=32, raw_mode=True)
+    item = ds[0]
+    assert set(item.keys()) == {"raw", "ref"}
+    assert item["raw"].dtype == torch.uint8
+    assert item["raw"].shape == (32, 32, 3)
+
+
+def test_train_engine_fast_requires_gpu():
+    import pytest
+    import torch
+
+    if torch.cuda.is_available():
+        pytest.skip("CPU-only check")
+    import train as train_cli
+
+    with pytest.raises(SystemExit):
+        train_cli.main(["--engine", "fast", "--epochs", "1",
+                        "--synthetic", "4", "--batch-size", "2",
+                        "--height", "32", "--width", "32"])

@@ -162,3 +162,130 @@ def test_stream_join_ordering():
     fresh = StreamJoin(cs)
     with pytest.raises(AssertionError):
         fresh.wait()  # wait() without mark() must fail loudly
+
+
+# ---------------------------------------------------------------------------
+# Round 2: unified fast engine (the train.py production path)
+# ---------------------------------------------------------------------------
+
+
+def test_fast_engine_loss_matches_eager_fp32():
+    """One eager-mode FastStepEngine step's loss vs the same computation in
+    plain PyTorch fp32 (same weights, same preprocessed inputs): validates
+    the full-NHWC wiring (build_inputs_u8 -> convs -> NormalizeNhwc -> VGG
+    -> Mse255) against the reference composition."""
+    import os
+
+    from waternet_amd.engine.fast import FastStepEngine
+    from waternet_amd.engine.losses import PERCEPTUAL_WEIGHT
+    from waternet_amd.engine.native import (
+        vgg_forward_nhwc,
+        waternet_forward_from_inputs,
+    )
+    from waternet_amd.models.vgg import normalize_imagenet
+    from waternet_amd.models.waternet import WaterNet
+    from waternet_amd.ops.functional import NormalizeNhwc, mse255_nhwc
+
+    torch.manual_seed(21)
+    model = WaterNet().to(DEV)
+    eng = FastStepEngine(model, batch_size=2, height=64, width=64,
+                         device=DEV, use_graph=False)
+    rng = np.random.default_rng(3)
+    raw = torch.from_numpy(
+        rng.integers(0, 256, size=(2, 64, 64, 3), dtype=np.uint8)).to(DEV)
+    ref = torch.from_numpy(
+        rng.integers(0, 256, size=(2, 64, 64, 3), dtype=np.uint8)).to(DEV)
+
+    inputs, ref_nhwc = eng._build_all_inputs(raw, ref)
+    with torch.no_grad():
+        out_nhwc = waternet_forward_from_inputs(model, *inputs)
+        fx = vgg_forward_nhwc(eng.vgg, NormalizeNhwc.apply(out_nhwc))
+        fy = vgg_forward_nhwc(eng.vgg, NormalizeNhwc.apply(ref_nhwc))
+        perceptual = mse255_nhwc(fx, fy, 512)
+        mse = mse255_nhwc(out_nhwc, ref_nhwc, 3)
+        loss_native = (PERCEPTUAL_WEIGHT * perceptual + mse).item()
+
+    # eager fp32 reference on the SAME preprocessed inputs
+    cmg_in = inputs[0]
+    raw_f = cmg_in[..., 0:3].permute(0, 3, 1, 2).float()
+    wb_f = cmg_in[..., 3:6].permute(0, 3, 1, 2).float()
+    he_f = cmg_in[..., 6:9].permute(0, 3, 1, 2).float()
+    gc_f = cmg_in[..., 9:12].permute(0, 3, 1, 2).float()
+    ref_f = ref_nhwc[..., :3].permute(0, 3, 1, 2).float()
+    os.environ["WATERNET_AMD_EAGER"] = "1"
+    try:
+        with torch.no_grad():
+            out_e = model(raw_f, wb_f, he_f, gc_f)
+            fx_e = eng.vgg(normalize_imagenet(out_e))
+            fy_e = eng.vgg(normalize_imagenet(ref_f))
+            dp = 255.0 * (fx_e - fy_e)
+            dm = 255.0 * (out_e - ref_f)
+            loss_eager = (PERCEPTUAL_WEIGHT * torch.mean(dp * dp)
+                          + torch.mean(dm * dm)).item()
+    finally:
+        os.environ.pop("WATERNET_AMD_EAGER", None)
+    rel = abs(loss_native - loss_eager) / max(abs(loss_eager), 1e-9)
+    assert rel < 0.05, (loss_native, loss_eager, rel)
+
+
+def test_fast_engine_ragged_batches():
+    """step_batch / eval_batch at a size different from the static batch
+    (the reference val loader's last batch is ragged, train.py:233-235)."""
+    from waternet_amd.engine.fast import FastStepEngine
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(2)
+    model = WaterNet().to(DEV)
+    eng = FastStepEngine(model, batch_size=4, height=64, width=64,
+                         device=DEV, use_graph=False)
+    rng = np.random.default_rng(1)
+
+    def batch(n):
+        return (torch.from_numpy(rng.integers(
+                    0, 256, size=(n, 64, 64, 3), dtype=np.uint8)).to(DEV),
+                torch.from_numpy(rng.integers(
+                    0, 256, size=(n, 64, 64, 3), dtype=np.uint8)).to(DEV))
+
+    raw, ref = batch(4)
+    eng.load_batch(raw, ref)
+    eng.step()
+    eng.step_batch(*batch(3))  # ragged train tail
+    eng.eval_batch(*batch(4))
+    eng.eval_batch(*batch(2))  # ragged val tail
+    torch.cuda.synchronize()
+    m = eng.metrics()
+    assert all(np.isfinite(v) for v in m.values()), m
+    ev = eng.eval_metrics()
+    assert all(np.isfinite(v) for v in ev.values()), ev
+    assert eng._eval_batches == 2
+
+
+def test_train_cli_fast_gpu(tmp_path):
+    """train.py end-to-end on GPU through the fast engine: synthetic data,
+    2 epochs, graph capture + ragged batches, reference-format outputs."""
+    import json
+    import os
+
+    import train as train_cli
+
+    os.environ["WATERNET_TRAINING_DIR"] = str(tmp_path)
+    try:
+        train_cli.main([
+            "--epochs", "2", "--batch-size", "8", "--height", "64",
+            "--width", "64", "--synthetic", "24", "--full-state",
+        ])
+    finally:
+        os.environ.pop("WATERNET_TRAINING_DIR", None)
+    savedir = tmp_path / "0"
+    sd = torch.load(savedir / "last.pt", map_location="cpu")
+    from waternet_amd.models.waternet import WaterNet
+
+    m = WaterNet()
+    m.load_state_dict(sd)  # byte-compatible 38-tensor schema
+    assert len(sd) == 38
+    train_csv = (savedir / "metrics-train.csv").read_text().splitlines()
+    assert train_csv[0] == "mse,ssim,psnr,perceptual_loss,loss"
+    assert len(train_csv) == 3  # header + 2 epochs
+    cfg = json.loads((savedir / "config.json").read_text())
+    assert cfg["batch_size"] == 8
+    assert (savedir / "last-trainstate.pt").exists()

@@ -392,3 +392,68 @@ def test_train_step_native_loss_decreases(ext):
         losses.append(loss.item())
     assert losses[-1] < losses[0], losses
     assert all(np.isfinite(v) for v in losses)
+
+
+# ---------------------------------------------------------------------------
+# Full-NHWC path kernels (round 2): uint8 input builders, in-layout
+# normalize, NHWC SSIM
+# ---------------------------------------------------------------------------
+
+
+def test_build_inputs_u8_parity(ext):
+    rng = np.random.default_rng(5)
+    mk = lambda: torch.from_numpy(  # noqa: E731
+        rng.integers(0, 256, size=(2, 12, 12, 3), dtype=np.uint8)).to(DEV)
+    raw, wb, ce, gc = mk(), mk(), mk(), mk()
+    cmg, rwb, rce, rgc = ext.build_inputs_u8(raw, wb, ce, gc)
+    f = lambda t: q(t.float() / 255.0)  # noqa: E731
+    exp_cmg = torch.cat([f(raw), f(wb), f(ce), f(gc)], dim=3)
+    assert torch.equal(cmg[..., :12].float(), exp_cmg)
+    assert cmg[..., 12:].abs().max().item() == 0
+    for rt, src in ((rwb, wb), (rce, ce), (rgc, gc)):
+        exp = torch.cat([f(raw), f(src)], dim=3)
+        assert torch.equal(rt[..., :6].float(), exp)
+        assert rt[..., 6:].abs().max().item() == 0
+
+
+def test_u8_to_nhwc(ext):
+    rng = np.random.default_rng(6)
+    u = torch.from_numpy(
+        rng.integers(0, 256, size=(2, 8, 8, 3), dtype=np.uint8)).to(DEV)
+    y = ext.u8_to_nhwc(u, 16)
+    assert y.shape == (2, 8, 8, 16)
+    assert torch.equal(y[..., :3].float(), q(u.float() / 255.0))
+    assert y[..., 3:].abs().max().item() == 0
+
+
+def test_normalize_nhwc_parity(ext):
+    from waternet_amd.models.vgg import IMAGENET_MEAN, IMAGENET_STD
+    from waternet_amd.ops.functional import NormalizeNhwc
+
+    torch.manual_seed(4)
+    x = to_nhwc_bf16(torch.rand(2, 3, 10, 10, device=DEV), 16)
+    x.requires_grad_(True)
+    y = NormalizeNhwc.apply(x)
+    mean = torch.tensor(IMAGENET_MEAN, device=DEV)
+    std = torch.tensor(IMAGENET_STD, device=DEV)
+    exp = (x.detach()[..., :3].float() - mean) / std
+    assert torch.allclose(y[..., :3].float(), exp, atol=2e-2), (
+        (y[..., :3].float() - exp).abs().max())
+    assert y[..., 3:].abs().max().item() == 0
+    # backward: dx = dy / std on logical channels
+    dy = to_nhwc_bf16(torch.rand(2, 3, 10, 10, device=DEV), 16)
+    y.backward(dy)
+    exp_dx = dy[..., :3].float() / std
+    assert torch.allclose(x.grad[..., :3].float(), exp_dx, atol=2e-2)
+
+
+def test_ssim_nhwc_parity(ext):
+    from waternet_amd.ops.ssim import ssim_nhwc
+    from waternet_amd.utils.metrics import _ssim_torch
+
+    torch.manual_seed(9)
+    a = torch.rand(2, 3, 32, 32, device=DEV)
+    b = (a * 0.7 + 0.2).clamp(0, 1)
+    got = ssim_nhwc(to_nhwc_bf16(a, 16), to_nhwc_bf16(b, 16), 3, 1.0)
+    want = _ssim_torch(q(a), q(b), 1.0, 11, 1.5, 0.01, 0.03)
+    assert torch.allclose(got, want.to(got.dtype), atol=2e-3), (got, want)

@@ -122,3 +122,43 @@ def test_torch_hub_local_load():
         out = model(rgb_t, wb_t, he_t, gc_t)
     arr = post(out)
     assert arr.shape == (1, 32, 32, 3) and arr.dtype == np.uint8
+
+
+def test_vgg_load_torchvision_schema():
+    """load_torchvision_state_dict against a state_dict with torchvision's
+    EXACT vgg19 key set (features.{0,2,5,7,10,12,14,16,19,21,23,25,28,30,
+    32,34}.{weight,bias} + classifier.{0,3,6}.*) — proves the loader maps
+    the real schema."""
+    import torch
+
+    from waternet_amd.models.vgg import PerceptualModel
+
+    conv_idx = [0, 2, 5, 7, 10, 12, 14, 16, 19, 21, 23, 25, 28, 30, 32, 34]
+    chans = [(64, 3), (64, 64), (128, 64), (128, 128),
+             (256, 128), (256, 256), (256, 256), (256, 256),
+             (512, 256), (512, 512), (512, 512), (512, 512),
+             (512, 512), (512, 512), (512, 512), (512, 512)]
+    g = torch.Generator().manual_seed(77)
+    sd = {}
+    for idx, (co, ci) in zip(conv_idx, chans):
+        sd[f"features.{idx}.weight"] = torch.randn(co, ci, 3, 3, generator=g)
+        sd[f"features.{idx}.bias"] = torch.randn(co, generator=g)
+    # classifier keys present in a full torchvision dict; must be ignored
+    sd["classifier.0.weight"] = torch.randn(1, 1)
+    sd["classifier.0.bias"] = torch.randn(1)
+    sd["classifier.3.weight"] = torch.randn(1, 1)
+    sd["classifier.3.bias"] = torch.randn(1)
+    sd["classifier.6.weight"] = torch.randn(1, 1)
+    sd["classifier.6.bias"] = torch.randn(1)
+
+    m = PerceptualModel(seed=0)
+    m.load_torchvision_state_dict(sd)
+    assert torch.equal(m.model[0].weight, sd["features.0.weight"])
+    assert torch.equal(m.model[34].weight, sd["features.34.weight"])
+    assert torch.equal(m.model[19].bias, sd["features.19.bias"])
+    # and the features-only subset form works too
+    m2 = PerceptualModel(seed=0)
+    feats_only = {k[len("features."):]: v for k, v in sd.items()
+                  if k.startswith("features.")}
+    m2.load_torchvision_state_dict(feats_only)
+    assert torch.equal(m2.model[34].weight, sd["features.34.weight"])

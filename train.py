@@ -4,16 +4,30 @@
 same outputs (last.pt state_dict checkpoint each epoch, metrics-train.csv,
 metrics-val.csv, config.json).
 
+Engines:
+  fast (default on a ROCm GPU) — the production step engine
+    (waternet_amd.engine.fast.FastStepEngine): GPU preprocess inside the
+    train step, native MFMA kernels end to end in NHWC bf16, fused Adam
+    flat arena, whole-step hipGraph replay, on-device metric accumulation
+    with ONE host sync per epoch. This is the same step bench.py measures —
+    training through this CLI runs at benchmark throughput.
+  eager — the PyTorch fp32 reference composition (CPU preprocess in the
+    dataloader, torch.optim.Adam), matching the reference loop shape
+    exactly (train.py:80-152). Used on CPU and for parity debugging.
+
 Extensions (do not break reference invocations):
   --data-root      dataset root containing raw-890/ and reference-890/
   --synthetic N    use N synthetic image pairs instead of UIEB on disk
   --num-workers    DataLoader workers (reference default: 0)
   --shuffle        shuffle the train loader (reference leaves it False)
   --full-state     also save optimizer/scheduler/epoch sidecar for resume
+  --engine         auto | fast | eager (auto = fast on GPU, eager on CPU)
+  --no-graph       disable hipGraph capture in the fast engine
 
 Multi-GPU: launch with `python -m torch.distributed.run --nproc-per-node N
 train.py ...` — one rank per GPU, RCCL all-reduce gradient sync
-(waternet_amd.parallel). Rank 0 writes checkpoints/metrics.
+(flat-arena collective inside the fast step / FlatBucketReducer in eager).
+Rank 0 writes checkpoints/metrics.
 """
 
 import argparse
@@ -32,11 +46,9 @@ from waternet_amd.engine.trainer import (
     eval_one_epoch,
     train_one_epoch,
 )
-from waternet_amd.models.vgg import PerceptualModel
 from waternet_amd.models.waternet import WaterNet
 from waternet_amd.parallel import (
     FlatBucketReducer,
-    distributed_env,
     init_distributed,
     shard_dataset,
 )
@@ -61,7 +73,80 @@ def parse_args(argv=None):
                         help="resume from a savedir holding last.pt (+ "
                              "last-trainstate.pt for optimizer/scheduler/"
                              "epoch state written by --full-state)")
+    parser.add_argument("--engine", choices=["auto", "fast", "eager"],
+                        default="auto",
+                        help="auto = fast native engine on GPU, eager "
+                             "PyTorch on CPU")
+    parser.add_argument("--no-graph", action="store_true",
+                        help="fast engine: disable hipGraph capture")
     return parser.parse_args(argv)
+
+
+def _select_engine(args, device):
+    if args.engine == "eager":
+        return False
+    fast_ok = (
+        device.type == "cuda"
+        and args.height % 8 == 0
+        and args.width % 8 == 0
+    )
+    if fast_ok:
+        from waternet_amd.ops import native_available
+
+        fast_ok = native_available()
+    if args.engine == "fast" and not fast_ok:
+        from waternet_amd.ops import native_load_error
+
+        raise SystemExit(
+            "--engine fast requires a ROCm GPU with the native extension "
+            f"and H,W divisible by 8 (load error: {native_load_error()})")
+    return fast_ok
+
+
+def _clean_state_dict(model):
+    """Independent CPU tensors (the fast engine's params are views into the
+    FusedAdam master arena; clone so last.pt matches the reference's
+    standalone-tensor checkpoint layout, train.py:308)."""
+    return {k: v.detach().cpu().clone()
+            for k, v in model.state_dict().items()}
+
+
+def _train_epoch_fast(engine, loader, batch_size, device, epoch_num,
+                      total_epochs, progress):
+    from waternet_amd.engine.fast import FastStepEngine  # noqa: F401
+
+    snap = engine.metrics_snapshot()
+    iterator = loader
+    if progress:
+        try:
+            from tqdm import tqdm
+
+            iterator = tqdm(loader, total=len(loader), ascii=True,
+                            desc=f"Epoch {epoch_num + 1}/{total_epochs}",
+                            bar_format="{l_bar}{bar:20}{r_bar}")
+        except ImportError:
+            pass
+    for batch in iterator:
+        raw, ref = batch["raw"], batch["ref"]
+        if raw.size(0) == batch_size:
+            engine.load_batch(raw, ref)
+            engine.step()
+        else:  # ragged tail batch: eager step at its true size
+            engine.step_batch(raw.to(device, non_blocking=True),
+                              ref.to(device, non_blocking=True))
+    m = engine.metrics_since(snap)  # ONE host sync per epoch
+    return {
+        "mse": m["mse255"], "ssim": m["ssim"], "psnr": m["psnr"],
+        "perceptual_loss": m["perceptual"], "loss": m["loss"],
+    }
+
+
+def _eval_epoch_fast(engine, loader, device):
+    engine.reset_eval()
+    for batch in loader:
+        engine.eval_batch(batch["raw"].to(device, non_blocking=True),
+                          batch["ref"].to(device, non_blocking=True))
+    return engine.eval_metrics()
 
 
 def main(argv=None):
@@ -86,16 +171,22 @@ def main(argv=None):
     else:
         device = torch.device("cpu")
 
+    use_fast = _select_engine(args, device)
+
     # Savedir: training/<max numbered subdir + 1>  (train.py:210-221)
     outputdir.mkdir(exist_ok=True)
     nums = [int(p.stem) for p in outputdir.glob("*")
             if p.is_dir() and p.stem.isdecimal()]
     savedir = outputdir / str(max(nums) + 1 if nums else 0)
 
-    # Dataset + split (800/90 on UIEB; proportional for synthetic)
+    # Dataset + split (800/90 on UIEB; proportional for synthetic).
+    # Fast engine: datasets yield uint8 HWC pairs; wb/gamma/clahe run
+    # on-GPU inside the step (the reference's CPU transform loop is its
+    # throughput bottleneck — SURVEY §6).
     if args.synthetic is not None:
         dataset = SyntheticUIEBDataset(
-            n_images=args.synthetic, im_height=args.height, im_width=args.width
+            n_images=args.synthetic, im_height=args.height,
+            im_width=args.width, raw_mode=use_fast,
         )
         n_val = max(1, int(0.1 * len(dataset)))
         split = [len(dataset) - n_val, n_val]
@@ -105,6 +196,7 @@ def main(argv=None):
             Path(args.data_root) / "reference-890",
             im_height=args.height,
             im_width=args.width,
+            raw_mode=use_fast,
         )
         split = [800, 90]
     train_dataset, val_dataset = torch.utils.data.random_split(dataset, split)
@@ -113,13 +205,17 @@ def main(argv=None):
     train_loader = torch.utils.data.DataLoader(
         train_dataset, batch_size=args.batch_size, shuffle=args.shuffle,
         num_workers=args.num_workers, pin_memory=device.type == "cuda",
+        persistent_workers=args.num_workers > 0,
     )
     val_loader = torch.utils.data.DataLoader(
-        val_dataset, batch_size=args.batch_size, num_workers=args.num_workers
+        val_dataset, batch_size=args.batch_size,
+        num_workers=args.num_workers,
+        pin_memory=device.type == "cuda",
     )
 
     if rank0:
-        print(f"Using device: {device}")
+        print(f"Using device: {device} "
+              f"({'fast native engine' if use_fast else 'eager engine'})")
 
     model = WaterNet()
     if args.resume is not None:
@@ -131,10 +227,28 @@ def main(argv=None):
     model.to(device)
     model.train()
 
-    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
-    scheduler = torch.optim.lr_scheduler.StepLR(
-        optimizer, step_size=10000, gamma=0.1
-    )
+    engine = None
+    reducer = None
+    if use_fast:
+        from waternet_amd.engine.fast import FastStepEngine
+
+        engine = FastStepEngine(
+            model, batch_size=args.batch_size, height=args.height,
+            width=args.width, device=device,
+            world_size=dist_env.world_size,
+            use_graph=not args.no_graph,
+        )
+        optimizer, scheduler = engine.opt, engine.sched
+        vgg_model = engine.vgg
+    else:
+        from waternet_amd.models.vgg import PerceptualModel
+
+        optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
+        scheduler = torch.optim.lr_scheduler.StepLR(
+            optimizer, step_size=10000, gamma=0.1
+        )
+        vgg_model = PerceptualModel().to(device).eval()
+
     start_epoch = 0
     if args.resume is not None:
         # full resume: optimizer/scheduler/epoch from the sidecar if present
@@ -143,31 +257,44 @@ def main(argv=None):
         sidecar = Path(args.resume) / "last-trainstate.pt"
         if sidecar.exists():
             state = torch.load(sidecar, map_location="cpu")
-            optimizer.load_state_dict(state["optimizer"])
-            scheduler.load_state_dict(state["scheduler"])
-            start_epoch = state["epoch"] + 1
-            if rank0:
-                print(f"Resumed epoch {start_epoch} from {sidecar}")
+            try:
+                optimizer.load_state_dict(state["optimizer"])
+                scheduler.load_state_dict(state["scheduler"])
+                start_epoch = state["epoch"] + 1
+                if rank0:
+                    print(f"Resumed epoch {start_epoch} from {sidecar}")
+            except (KeyError, ValueError) as e:
+                if rank0:
+                    print(f"WARNING: sidecar optimizer state incompatible "
+                          f"with this engine ({e!r}); optimizer state reset")
 
-    vgg_model = PerceptualModel().to(device).eval()
-
-    reducer = None
     if dist_env.world_size > 1:
-        # Broadcast initial params so all ranks start identical, then
-        # all-reduce gradients each step through one flat bucket.
-        reducer = FlatBucketReducer(model, dist_env)
-        reducer.broadcast_params()
+        if use_fast:
+            # all ranks start from rank0's flat master arena (one-time)
+            torch.distributed.broadcast(engine.opt.master, src=0)
+        else:
+            # Broadcast initial params so all ranks start identical, then
+            # all-reduce gradients each step through one flat bucket.
+            reducer = FlatBucketReducer(model, dist_env)
+            reducer.broadcast_params()
 
     saved_train = {k: [] for k in TRAIN_METRICS_NAMES}
     saved_val = {k: [] for k in VAL_METRICS_NAMES}
 
     for epoch in range(start_epoch, args.epochs):
-        train_metrics = train_one_epoch(
-            model, train_loader, optimizer, scheduler, vgg_model, device,
-            epoch_num=epoch, total_epochs=args.epochs,
-            grad_reducer=reducer, progress=rank0,
-        )
-        val_metrics = eval_one_epoch(model, val_loader, device, vgg_model)
+        if use_fast:
+            train_metrics = _train_epoch_fast(
+                engine, train_loader, args.batch_size, device,
+                epoch_num=epoch, total_epochs=args.epochs, progress=rank0,
+            )
+            val_metrics = _eval_epoch_fast(engine, val_loader, device)
+        else:
+            train_metrics = train_one_epoch(
+                model, train_loader, optimizer, scheduler, vgg_model, device,
+                epoch_num=epoch, total_epochs=args.epochs,
+                grad_reducer=reducer, progress=rank0,
+            )
+            val_metrics = eval_one_epoch(model, val_loader, device, vgg_model)
 
         if dist_env.world_size > 1:
             train_metrics = dist_env.average_metrics(train_metrics)
@@ -187,7 +314,7 @@ def main(argv=None):
 
         if rank0:
             savedir.mkdir(exist_ok=True)
-            torch.save(model.state_dict(), savedir / "last.pt")
+            torch.save(_clean_state_dict(model), savedir / "last.pt")
             if args.full_state:
                 torch.save(
                     {

@@ -39,6 +39,13 @@ at::Tensor sqdiff255_bwd(const at::Tensor& a, const at::Tensor& b,
                          const at::Tensor& gscale, double sign);
 at::Tensor out_to_u8(const at::Tensor& x);
 at::Tensor u8_to_nchw(const at::Tensor& x);
+std::vector<at::Tensor> build_inputs_u8(const at::Tensor& raw,
+                                        const at::Tensor& wb,
+                                        const at::Tensor& ce,
+                                        const at::Tensor& gc);
+at::Tensor u8_to_nhwc(const at::Tensor& x, int64_t Cp);
+at::Tensor normalize_nhwc_fwd(const at::Tensor& x);
+at::Tensor normalize_nhwc_bwd(const at::Tensor& dy);
 void adam_step(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
                at::Tensor& v, const at::Tensor& lr_buf, double b1, double b2,
                double eps, at::Tensor& step_buf);
@@ -51,6 +58,9 @@ at::Tensor maxpool2x2_bwd(const at::Tensor& dy, const at::Tensor& idx,
 // ssim.hip
 at::Tensor ssim_sum(const at::Tensor& a, const at::Tensor& b,
                     double data_range, double k1, double k2);
+at::Tensor ssim_sum_nhwc(const at::Tensor& a, const at::Tensor& b,
+                         int64_t Clog, double data_range, double k1,
+                         double k2);
 
 // preprocess.hip
 std::vector<at::Tensor> preprocess_all(const at::Tensor& raw_u8);
@@ -78,8 +88,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step);
   m.def("out_to_u8", &out_to_u8);
   m.def("u8_to_nchw", &u8_to_nchw);
+  m.def("build_inputs_u8", &build_inputs_u8,
+        "uint8 HWC batch -> cat-folded NHWC bf16 conv inputs");
+  m.def("u8_to_nhwc", &u8_to_nhwc);
+  m.def("normalize_nhwc_fwd", &normalize_nhwc_fwd);
+  m.def("normalize_nhwc_bwd", &normalize_nhwc_bwd);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd);
   m.def("ssim_sum", &ssim_sum);
+  m.def("ssim_sum_nhwc", &ssim_sum_nhwc);
   m.def("preprocess_all", &preprocess_all);
 }

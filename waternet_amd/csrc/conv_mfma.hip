@@ -49,7 +49,12 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
     bf16_t* __restrict__ Y,         // (N, H, W, Kp)
     int N, int H, int W, int Cp, int log2Cp, int Kp, int Klog, int act,
     const bf16_t* __restrict__ Zero16,    // 16 B of zeros (halo/pad source)
-    float* __restrict__ Y32 = nullptr) {  // SPLITK: fp32 partials (pre-zeroed)
+    float* __restrict__ Y32 = nullptr) {  // SPLITK: fp32 partial slabs.
+                                        // NOT pre-zeroed (at::empty): every
+                                        // slice block fully stores its m<M
+                                        // rows, and finalize only reads
+                                        // i < M*Kp — an epilogue edit that
+                                        // skips stores must zero-fill first.
   constexpr int BM = 128;
   constexpr int PAD = KS / 2;
   constexpr int RS = KS * KS;

@@ -59,6 +59,61 @@ __global__ void k_build_inputs(const float* __restrict__ raw,
   }
 }
 
+// uint8 variant: builds the same four conv inputs straight from the uint8
+// HWC batch tensors (raw + the GPU-preprocess outputs), /255 — removes the
+// per-tensor u8->NCHW-fp32 bridges and the NCHW intermediates entirely
+// (full-NHWC training path; reference arr2ten training_utils.py:11-27 +
+// torch.cat net.py:46/76 in ONE kernel).
+__global__ void k_build_inputs_u8(const uint8_t* __restrict__ raw,
+                                  const uint8_t* __restrict__ wb,
+                                  const uint8_t* __restrict__ ce,
+                                  const uint8_t* __restrict__ gc,
+                                  bf16_t* __restrict__ cmg_in,
+                                  bf16_t* __restrict__ rwb_in,
+                                  bf16_t* __restrict__ rce_in,
+                                  bf16_t* __restrict__ rgc_in, long NHW) {
+  constexpr float S = 1.f / 255.f;
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const long s = p * 3;
+    float r0 = raw[s] * S, r1 = raw[s + 1] * S, r2 = raw[s + 2] * S;
+    float w0 = wb[s] * S, w1 = wb[s + 1] * S, w2 = wb[s + 2] * S;
+    float c0 = ce[s] * S, c1 = ce[s + 1] * S, c2 = ce[s + 2] * S;
+    float g0 = gc[s] * S, g1 = gc[s + 1] * S, g2 = gc[s + 2] * S;
+    bf16_t* o = cmg_in + p * 16;
+    o[0] = f2bf(r0); o[1] = f2bf(r1); o[2] = f2bf(r2);
+    o[3] = f2bf(w0); o[4] = f2bf(w1); o[5] = f2bf(w2);
+    o[6] = f2bf(c0); o[7] = f2bf(c1); o[8] = f2bf(c2);
+    o[9] = f2bf(g0); o[10] = f2bf(g1); o[11] = f2bf(g2);
+    o[12] = o[13] = o[14] = o[15] = f2bf(0.f);
+    auto emit = [&](bf16_t* dst, float a0, float a1, float a2) {
+      bf16_t* q = dst + p * 16;
+      q[0] = f2bf(r0); q[1] = f2bf(r1); q[2] = f2bf(r2);
+      q[3] = f2bf(a0); q[4] = f2bf(a1); q[5] = f2bf(a2);
+#pragma unroll
+      for (int i = 6; i < 16; ++i) q[i] = f2bf(0.f);
+    };
+    emit(rwb_in, w0, w1, w2);
+    emit(rce_in, c0, c1, c2);
+    emit(rgc_in, g0, g1, g2);
+  }
+}
+
+// uint8 HWC -> (N,H,W,Cp) bf16 [0,1] (the ref tensor for the NHWC loss path)
+__global__ void k_u8_to_nhwc(const uint8_t* __restrict__ x,
+                             bf16_t* __restrict__ y, long NHW, int Cp) {
+  constexpr float S = 1.f / 255.f;
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const uint8_t* s = x + p * 3;
+    bf16_t* d = y + (long)p * Cp;
+    d[0] = f2bf(s[0] * S);
+    d[1] = f2bf(s[1] * S);
+    d[2] = f2bf(s[2] * S);
+    for (int c = 3; c < Cp; ++c) d[c] = f2bf(0.f);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // NCHW fp32 <-> NHWC bf16 bridges
 // ---------------------------------------------------------------------------
@@ -253,6 +308,36 @@ __global__ void k_normalize_vgg_bwd(const bf16_t* __restrict__ dy,
   }
 }
 
+// NHWC bf16 -> NHWC bf16 in-layout normalize (full-NHWC loss path: the
+// WaterNet output and ref stay NHWC from the fusion kernel through the VGG
+// towers — no NCHW round trip).
+__global__ void k_normalize_nhwc_fwd(const bf16_t* __restrict__ x,
+                                     bf16_t* __restrict__ y, long NHW,
+                                     int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const bf16_t* s = x + (long)p * Cp;
+    bf16_t* d = y + (long)p * Cp;
+#pragma unroll
+    for (int c = 0; c < 3; ++c)
+      d[c] = f2bf((bf2f(s[c]) - IMNET_MEAN[c]) / IMNET_STD[c]);
+    for (int c = 3; c < Cp; ++c) d[c] = f2bf(0.f);
+  }
+}
+
+__global__ void k_normalize_nhwc_bwd(const bf16_t* __restrict__ dy,
+                                     bf16_t* __restrict__ dx, long NHW,
+                                     int Cp) {
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < NHW;
+       p += (long)gridDim.x * blockDim.x) {
+    const bf16_t* s = dy + (long)p * Cp;
+    bf16_t* d = dx + (long)p * Cp;
+#pragma unroll
+    for (int c = 0; c < 3; ++c) d[c] = f2bf(bf2f(s[c]) / IMNET_STD[c]);
+    for (int c = 3; c < Cp; ++c) d[c] = f2bf(0.f);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Scaled squared-difference reduction (K18/K19):
 //   sum over logical elements of (255*(a-b))^2, fp64 accumulator.
@@ -392,6 +477,67 @@ std::vector<at::Tensor> build_inputs(const at::Tensor& raw,
                      HW);
   HIP_CHECK_LAST();
   return {cmg_in, rwb, rce, rgc};
+}
+
+std::vector<at::Tensor> build_inputs_u8(const at::Tensor& raw,
+                                        const at::Tensor& wb,
+                                        const at::Tensor& ce,
+                                        const at::Tensor& gc) {
+  TORCH_CHECK(raw.is_cuda() && raw.dtype() == at::kByte && raw.dim() == 4 &&
+              raw.size(3) == 3, "raw must be (N,H,W,3) uint8 CUDA");
+  const long N = raw.size(0), H = raw.size(1), W = raw.size(2);
+  const long NHW = N * H * W;
+  auto opts = raw.options().dtype(at::kBFloat16);
+  auto cmg_in = at::empty({N, H, W, 16}, opts);
+  auto rwb = at::empty({N, H, W, 16}, opts);
+  auto rce = at::empty({N, H, W, 16}, opts);
+  auto rgc = at::empty({N, H, W, 16}, opts);
+  hipLaunchKernelGGL(k_build_inputs_u8, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), raw.contiguous().data_ptr<uint8_t>(),
+                     wb.contiguous().data_ptr<uint8_t>(),
+                     ce.contiguous().data_ptr<uint8_t>(),
+                     gc.contiguous().data_ptr<uint8_t>(),
+                     (bf16_t*)cmg_in.data_ptr(), (bf16_t*)rwb.data_ptr(),
+                     (bf16_t*)rce.data_ptr(), (bf16_t*)rgc.data_ptr(), NHW);
+  HIP_CHECK_LAST();
+  return {cmg_in, rwb, rce, rgc};
+}
+
+at::Tensor u8_to_nhwc(const at::Tensor& x, int64_t Cp) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kByte && x.dim() == 4 &&
+              x.size(3) == 3);
+  const long N = x.size(0), H = x.size(1), W = x.size(2);
+  auto y = at::empty({N, H, W, Cp}, x.options().dtype(at::kBFloat16));
+  const long NHW = N * H * W;
+  hipLaunchKernelGGL(k_u8_to_nhwc, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), x.contiguous().data_ptr<uint8_t>(),
+                     (bf16_t*)y.data_ptr(), NHW, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor normalize_nhwc_fwd(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kBFloat16 && x.dim() == 4);
+  const long Cp = x.size(3);
+  const long NHW = x.numel() / Cp;
+  auto y = at::empty_like(x);
+  hipLaunchKernelGGL(k_normalize_nhwc_fwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)x.contiguous().data_ptr(),
+                     (bf16_t*)y.data_ptr(), NHW, (int)Cp);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor normalize_nhwc_bwd(const at::Tensor& dy) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == at::kBFloat16 && dy.dim() == 4);
+  const long Cp = dy.size(3);
+  const long NHW = dy.numel() / Cp;
+  auto dx = at::empty_like(dy);
+  hipLaunchKernelGGL(k_normalize_nhwc_bwd, dim3(grid1d(NHW)), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)dy.contiguous().data_ptr(),
+                     (bf16_t*)dx.data_ptr(), NHW, (int)Cp);
+  HIP_CHECK_LAST();
+  return dx;
 }
 
 at::Tensor nchw_to_nhwc(const at::Tensor& x, int64_t Cp) {

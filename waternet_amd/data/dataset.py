@@ -49,7 +49,11 @@ class UIEBDataset(torch.utils.data.Dataset):
         im_height: Optional[int] = None,
         im_width: Optional[int] = None,
         transform=None,
+        raw_mode: bool = False,
     ):
+        """raw_mode=True: skip the CPU preprocess transforms and return
+        {raw, ref} uint8 HWC tensors — the fast GPU-preprocess train path
+        runs wb/gamma/clahe on-device inside the step instead."""
         raw_im_fns = sorted(
             p.name for p in Path(raw_dir).glob("*.png")
         )
@@ -59,6 +63,7 @@ class UIEBDataset(torch.utils.data.Dataset):
         assert set(raw_im_fns) == set(ref_im_fns), "raw/ref file mismatch"
 
         self.transform = transform if transform is not None else PairedAugment()
+        self.raw_mode = raw_mode
         self.raw_dir = Path(raw_dir)
         self.ref_dir = Path(ref_dir)
         self.im_fns = raw_im_fns
@@ -86,6 +91,12 @@ class UIEBDataset(torch.utils.data.Dataset):
         if self.transform is not None:
             raw_im, ref_im = self.transform(image=raw_im, mask=ref_im)
 
+        if self.raw_mode:
+            return {
+                "raw": torch.from_numpy(np.ascontiguousarray(raw_im)),
+                "ref": torch.from_numpy(np.ascontiguousarray(ref_im)),
+            }
+
         wb, gc, he = preprocess_transform(raw_im)
 
         return {
@@ -107,12 +118,13 @@ class SyntheticUIEBDataset(torch.utils.data.Dataset):
     """
 
     def __init__(self, n_images=800, im_height=112, im_width=112, seed=0,
-                 run_transforms=True):
+                 run_transforms=True, raw_mode=False):
         self.n = n_images
         self.h = im_height
         self.w = im_width
         self.seed = seed
         self.run_transforms = run_transforms
+        self.raw_mode = raw_mode
 
     def __len__(self):
         return self.n
@@ -125,6 +137,9 @@ class SyntheticUIEBDataset(torch.utils.data.Dataset):
 
     def __getitem__(self, idx):
         raw_im, ref_im = self.raw_uint8(idx)
+        if self.raw_mode:
+            return {"raw": torch.from_numpy(raw_im),
+                    "ref": torch.from_numpy(ref_im)}
         out = {"raw": arr2ten(raw_im), "ref": arr2ten(ref_im)}
         if self.run_transforms:
             wb, gc, he = preprocess_transform(raw_im)

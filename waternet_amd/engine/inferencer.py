@@ -33,16 +33,16 @@ class InferenceEngine:
 
     @torch.no_grad()
     def _body(self):
+        from waternet_amd.engine.native import waternet_forward_from_inputs
+
         raw = self.raw_static
         wb, gc, he = gpu_transform_batch(raw)
         e = ext()
-        raw_f = e.u8_to_nchw(raw)
-        wb_f = e.u8_to_nchw(wb)
-        gc_f = e.u8_to_nchw(gc)
-        he_f = e.u8_to_nchw(he)
-        out = self.model(raw_f, wb_f, he_f, gc_f)  # he fills the ce slot
-        # out is NCHW fp32; fused u8 postprocess expects NHWC bf16 — convert
-        out_nhwc = e.nchw_to_nhwc(out, 16)
+        # he fills the ce slot (reference inference.py:191 -> net.py:99);
+        # full-NHWC: fused uint8 cat-fold -> convs -> fused u8 postprocess,
+        # no NCHW tensor anywhere in the frame pipeline.
+        inputs = e.build_inputs_u8(raw, wb, he, gc)
+        out_nhwc = waternet_forward_from_inputs(self.model, *inputs)
         self.out_static = e.out_to_u8(out_nhwc)
 
     def _capture(self):

@@ -54,18 +54,12 @@ def _state(model) -> WaterNetNativeState:
     return st
 
 
-def waternet_forward_native(model, x, wb, ce, gc):
-    """x, wb, ce, gc: (N,3,H,W) fp32 CUDA -> (N,3,H,W) fp32.
-
-    Kernel pipeline: build_inputs (fused cat, net.py:46/76) -> 8 CMG convs
-    -> 3x3 refiner convs -> gated fusion (net.py:104-108) -> NHWC->NCHW."""
+def waternet_forward_from_inputs(model, cmg_in, rwb_in, rce_in, rgc_in):
+    """Core NHWC pipeline from pre-built (cat-folded) conv inputs:
+    8 CMG convs -> 3x3 refiner convs -> gated fusion (net.py:104-108).
+    Inputs/output (N,H,W,16) bf16, 3 logical output channels — the
+    full-NHWC training/inference paths stay in this layout end to end."""
     st = _state(model)
-    xc = x.float().contiguous()
-    inputs = ext().build_inputs(xc, wb.float().contiguous(),
-                                ce.float().contiguous(),
-                                gc.float().contiguous())
-    cmg_in, rwb_in, rce_in, rgc_in = inputs
-
     t = cmg_in
     for spec in st.cmg_specs:
         t = conv_bias_act(t, spec)
@@ -79,7 +73,19 @@ def waternet_forward_native(model, x, wb, ce, gc):
             r = conv_bias_act(r, spec)
         refined.append(r)
 
-    out_nhwc = GatedFusion.apply(maps, refined[0], refined[1], refined[2])
+    return GatedFusion.apply(maps, refined[0], refined[1], refined[2])
+
+
+def waternet_forward_native(model, x, wb, ce, gc):
+    """x, wb, ce, gc: (N,3,H,W) fp32 CUDA -> (N,3,H,W) fp32.
+
+    Kernel pipeline: build_inputs (fused cat, net.py:46/76) ->
+    waternet_forward_from_inputs -> NHWC->NCHW (public NCHW contract)."""
+    xc = x.float().contiguous()
+    inputs = ext().build_inputs(xc, wb.float().contiguous(),
+                                ce.float().contiguous(),
+                                gc.float().contiguous())
+    out_nhwc = waternet_forward_from_inputs(model, *inputs)
     return NhwcToNchw.apply(out_nhwc, 3)
 
 
@@ -102,17 +108,40 @@ class VggNativeState:
                 spec.mark_dirty()
 
 
-def vgg_forward_native(vgg_model, x):
-    """x: (N,3,H,W) fp32 (already ImageNet-normalized by the caller) ->
-    (N,512,H/16,W/16) fp32."""
+def vgg_state(vgg_model) -> VggNativeState:
     st = getattr(vgg_model, "_wn_native_state", None)
     if st is None:
         st = VggNativeState(vgg_model)
         vgg_model._wn_native_state = st
-    t = NchwToNhwc.apply(x, 16)
-    for kind, spec in st.plan:
+    return st
+
+
+def vgg_forward_nhwc(vgg_model, x_nhwc):
+    """(N,H,W,16) bf16 (already ImageNet-normalized) ->
+    (N,H/16,W/16,512) bf16 — the full-NHWC perceptual tower."""
+    t = x_nhwc
+    for kind, spec in vgg_state(vgg_model).plan:
         if kind == "conv":
             t = conv_bias_act(t, spec)
         else:
             t = MaxPool2x2.apply(t)
+    return t
+
+
+def vgg_prepack(vgg_model):
+    """Force-build the ConvSpec table and pack all 16 VGG conv weights ON
+    THE CURRENT STREAM. The frozen VGG never repacks afterwards, so a step
+    engine that runs the fy tower on a side stream must call this once from
+    the main stream (otherwise the first step's fx tower reads buffers
+    packed on the side stream without ordering)."""
+    for kind, spec in vgg_state(vgg_model).plan:
+        if kind == "conv":
+            spec.refresh_if_needed()
+
+
+def vgg_forward_native(vgg_model, x):
+    """x: (N,3,H,W) fp32 (already ImageNet-normalized by the caller) ->
+    (N,512,H/16,W/16) fp32."""
+    t = NchwToNhwc.apply(x, 16)
+    t = vgg_forward_nhwc(vgg_model, t)
     return NhwcToNchw.apply(t, 512)

@@ -52,8 +52,15 @@ class ConvSpec:
         ver = (w._version, w.data_ptr())
         if self._wp is None or ver != self._version:
             wc = w.data.contiguous()
-            self._wp = ext().pack_weight_fwd(wc, self.Kp, self.Cp)
-            self._wd = ext().pack_weight_dgrad(wc, self.Kp, self.Cp)
+            if self._wp is None:
+                self._wp = ext().pack_weight_fwd(wc, self.Kp, self.Cp)
+                self._wd = ext().pack_weight_dgrad(wc, self.Kp, self.Cp)
+            else:
+                # Rewrite IN PLACE: pack_all descriptors bake these device
+                # pointers, so the buffers must stay at the same address
+                # for the life of the spec (never reallocate).
+                self._wp.copy_(ext().pack_weight_fwd(wc, self.Kp, self.Cp))
+                self._wd.copy_(ext().pack_weight_dgrad(wc, self.Kp, self.Cp))
             self._version = ver
 
     def mark_dirty(self):

@@ -62,6 +62,19 @@ class MaxPool2x2(torch.autograd.Function):
         return ext().maxpool2x2_bwd(dy.contiguous(), idx, *ctx.hw)
 
 
+class NormalizeNhwc(torch.autograd.Function):
+    """ImageNet normalize IN the NHWC bf16 layout (full-NHWC loss path:
+    WaterNet output -> VGG input with no NCHW round trip)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return ext().normalize_nhwc_fwd(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        return ext().normalize_nhwc_bwd(dy.contiguous())
+
+
 class NormalizeVgg(torch.autograd.Function):
     """ImageNet normalize fused with NCHW fp32 -> NHWC bf16 (Cp=16)."""
 

@@ -5,6 +5,17 @@ import torch
 from waternet_amd.ops import ext
 
 
+def ssim_nhwc(preds, target, clog, data_range, k1=0.01, k2=0.03):
+    """SSIM over NHWC bf16 tensors (clog logical channels) — the full-NHWC
+    metric path (11x11 gaussian, sigma 1.5, torchmetrics semantics)."""
+    a = preds.contiguous()
+    b = target.contiguous()
+    s = ext().ssim_sum_nhwc(a, b, clog, float(data_range), k1, k2)
+    n, h, w, _ = a.shape
+    count = n * clog * (h - 10) * (w - 10)
+    return (s / count).to(torch.float32)
+
+
 def ssim_native(preds, target, data_range, kernel_size=11, sigma=1.5,
                 k1=0.01, k2=0.03):
     if kernel_size != 11 or abs(sigma - 1.5) > 1e-9:
