@@ -66,7 +66,11 @@ def parse_args(argv=None):
     parser.add_argument("--data-root", type=str, default="data")
     parser.add_argument("--synthetic", type=int, default=None, metavar="N",
                         help="use N synthetic image pairs instead of UIEB")
-    parser.add_argument("--num-workers", type=int, default=0)
+    parser.add_argument("--num-workers", type=int, default=None,
+                        help="DataLoader workers; default 0 on the eager "
+                             "engine (reference behavior) and 4 on the fast "
+                             "engine (decode/collate hides under the GPU "
+                             "step)")
     parser.add_argument("--shuffle", action="store_true")
     parser.add_argument("--full-state", action="store_true")
     parser.add_argument("--resume", type=str, default=None, metavar="DIR",
@@ -202,15 +206,20 @@ def main(argv=None):
     train_dataset, val_dataset = torch.utils.data.random_split(dataset, split)
     train_dataset = shard_dataset(train_dataset, dist_env)
 
+    num_workers = args.num_workers
+    if num_workers is None:
+        num_workers = min(4, os.cpu_count() or 1) if use_fast else 0
+
     train_loader = torch.utils.data.DataLoader(
         train_dataset, batch_size=args.batch_size, shuffle=args.shuffle,
-        num_workers=args.num_workers, pin_memory=device.type == "cuda",
-        persistent_workers=args.num_workers > 0,
+        num_workers=num_workers, pin_memory=device.type == "cuda",
+        persistent_workers=num_workers > 0,
     )
     val_loader = torch.utils.data.DataLoader(
         val_dataset, batch_size=args.batch_size,
-        num_workers=args.num_workers,
+        num_workers=num_workers,
         pin_memory=device.type == "cuda",
+        persistent_workers=num_workers > 0,
     )
 
     if rank0:
@@ -236,7 +245,9 @@ def main(argv=None):
             model, batch_size=args.batch_size, height=args.height,
             width=args.width, device=device,
             world_size=dist_env.world_size,
-            use_graph=not args.no_graph,
+            # keep the RCCL collective out of graph capture under DDP
+            # (same policy as bench.py)
+            use_graph=(not args.no_graph) and dist_env.world_size == 1,
         )
         optimizer, scheduler = engine.opt, engine.sched
         vgg_model = engine.vgg
