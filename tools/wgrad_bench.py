@@ -2,12 +2,16 @@
 WN_WGRAD_V4=0|1). Shapes = the WaterNet training layers at bs=16 112^2."""
 
 import os
+import sys
 import time
+from pathlib import Path
 
 import torch
 
-from waternet_amd.ops import ext
-from waternet_amd.ops.conv import pow2_channels
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from waternet_amd.ops import ext  # noqa: E402
+from waternet_amd.ops.conv import pow2_channels  # noqa: E402
 
 SHAPES = [  # (ks, C, K, label)
     (7, 12, 128, "cmg.conv1"),

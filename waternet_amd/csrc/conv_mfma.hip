@@ -1058,8 +1058,13 @@ __global__ __launch_bounds__(512, 2) void k_conv_wgrad8(
   const int aDst = aKblk * KBS_H + aMblk * TR_MBS + aR * 8;
 
   // ---- B slots (X gather) ----
-  int bMoff[SB], bC_[SB], bDy[SB], bDx[SB], bDst[SB];
-  bool bPad[SB];
+  // NOTE fixed [3] bound (SB <= 3 for every instantiation): sizing these
+  // with the template-dependent SB makes this toolchain's HOST pass
+  // silently fail to emit the kernel's __device_stub__ when the array is
+  // captured by the stage lambda (undefined symbol at dlopen).
+  static_assert(SB <= 3, "bump the fixed B-slot array bound");
+  int bMoff[3], bC_[3], bDy[3], bDx[3], bDst[3];
+  bool bPad[3];
 #pragma unroll
   for (int s = 0; s < SB; ++s) {
     const int u = tid + s * 512;
@@ -1153,7 +1158,10 @@ __global__ __launch_bounds__(512, 2) void k_conv_wgrad8(
 
   for (long j = 0; j < nHalf; ++j) {
     // own pieces for half j landed (j+1, j+2 may stay in flight — T4)
-    wait_halves((int)min((long)2, nHalf - 1 - j));
+    {
+      const long r_ = nHalf - 1 - j;  // no min(long,long): host pass lacks
+      wait_halves(r_ > 2 ? 2 : (int)r_);  // the overload (silent stub kill)
+    }
     __builtin_amdgcn_s_barrier();
     if (j + 3 < nHalf) stage(j + 3);  // into slot (j-1)&3: reads done by all
                                       // waves before they arrived above
@@ -1239,6 +1247,43 @@ __global__ __launch_bounds__(512, 2) void k_conv_wgrad8(
     }
   }
 }
+
+template __global__ void k_conv_wgrad8<7, 128, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<5, 128, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<3, 128, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<7, 64, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<5, 64, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<3, 64, 384>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<1, 64, 128>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<7, 32, 256>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
+template __global__ void k_conv_wgrad8<5, 32, 256>(
+    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
+    int, int, int, int, unsigned long long, unsigned long long,
+    const bf16_t*);
 
 // ---------------------------------------------------------------------------
 // Small-K weight gradient (K <= 4: the 64->3 / 32->3 output convs).
