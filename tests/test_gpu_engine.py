@@ -281,8 +281,8 @@ def test_train_cli_fast_gpu(tmp_path):
     from waternet_amd.models.waternet import WaterNet
 
     m = WaterNet()
-    m.load_state_dict(sd)  # byte-compatible 38-tensor schema
-    assert len(sd) == 38
+    m.load_state_dict(sd)  # byte-compatible 34-tensor schema
+    assert len(sd) == 34
     train_csv = (savedir / "metrics-train.csv").read_text().splitlines()
     assert train_csv[0] == "mse,ssim,psnr,perceptual_loss,loss"
     assert len(train_csv) == 3  # header + 2 epochs

@@ -47,7 +47,7 @@ def test_forward_resolution_agnostic():
 
 
 def test_state_dict_schema():
-    """Checkpoint contract: 38 tensors, exact reference key names
+    """Checkpoint contract: 34 tensors, exact reference key names
     (train.py:308)."""
     model = WaterNet()
     sd = model.state_dict()

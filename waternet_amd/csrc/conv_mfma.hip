@@ -1815,6 +1815,11 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
       const int gy4 = (KG + BRV - 1) / BRV;
       int split4 = std::max(1, 512 / std::max(1, gx4 * gy4));
       split4 = (int)std::min<long>(split4, nCh4);
+      static const int split_cap = [] {  // debug: WN_WGRAD_SPLIT caps z
+        const char* e = getenv("WN_WGRAD_SPLIT");
+        return e ? atoi(e) : 0;
+      }();
+      if (split_cap > 0) split4 = std::min(split4, split_cap);
       const size_t lds4 =
           (size_t)4 * ((BKV / 16 + BRV / 16) * (8 * TR_MBS + 16)) *
           sizeof(bf16_t);

@@ -11,7 +11,7 @@ Architecture and state_dict schema replicate /root/reference/waternet/net.py:
 Parameters are stored as standard nn.Conv2d modules named exactly as the
 reference so checkpoints interchange bit-for-bit:
   cmg.conv{1..8}.{weight,bias}, {wb,ce,gc}_refiner.conv{1..3}.{weight,bias}
-(38 tensors, 1,090,668 params — train.py:308 schema).
+(34 tensors, 1,090,668 params — train.py:308 schema).
 
 Execution: on a ROCm GPU the forward runs through the hand-written CDNA4 HIP
 engine (NHWC bf16 MFMA implicit-GEMM convolutions with fused bias+ReLU/
