@@ -984,308 +984,6 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
 }
 
 // ---------------------------------------------------------------------------
-// v4 wgrad: 8-wave (512-thread) big-tile phase pipeline (guide §5 T3+T4+T5).
-//
-// The v3 kernel above is DMA-issue/landing bound at ~25% MFMA duty: per
-// 64-m chunk each 4-wave block stages (BK+BR)*64 elems for BK*BR*64 MACs
-// and fully drains the DMA at the per-chunk __syncthreads (2-phase
-// structure — the guide's "stage+vmcnt+barrier is 72% of the critical
-// path" regime). v4 changes the economics and the schedule:
-//   - 512 threads, BKxBR up to 128x384 -> up to 1.9x the MACs per staged
-//     byte, and r==8 LDS pad chunks are never staged (v3 issued a zero
-//     glds for every pad slot: ~12% of all staging issues).
-//   - m-chunks staged in HALVES (32 m) into a 4-slot LDS ring with
-//     COUNTED s_waitcnt vmcnt (2 halves stay in flight across barriers,
-//     never drained to 0 in the loop — T4) and ONE raw s_barrier per half.
-//   - per-kstep counted-lgkm tr-read interleave (2-deep ring) as v3.
-//   - s_setprio(1) around each MFMA cluster (T5 — the phase split creates
-//     the {staging vs MFMA} wave role diversity it needs).
-// Same operand layout contract as v3 (TR_MBS subtiles, ds_read_b64_tr_b16
-// fragments, KMAP-0), same fp32 atomicAdd epilogue.
-// ---------------------------------------------------------------------------
-
-constexpr int KBS_H = 8 * TR_MBS + 16;  // per-half kblk stride (296 dw =
-                                        // 8 mod 32: conflict-free staging)
-
-template <int KS, int BK, int BR>
-__global__ __launch_bounds__(512, 2) void k_conv_wgrad8(
-    const bf16_t* __restrict__ dY,  // (N,H,W,Kp)
-    const bf16_t* __restrict__ X,   // (N,H,W,Cp)
-    float* __restrict__ dW,         // (K, C, KS, KS) fp32, pre-zeroed
-    int N, int H, int W, int Cp, int log2Cp, int Kp, int K, int C,
-    int splitm, unsigned long long mulHW, unsigned long long mulW,
-    const bf16_t* __restrict__ Zero16w) {
-  static_assert(WN_MFMA_KMAP == 0, "wgrad staging assumes KMAP 0");
-  constexpr int PAD = KS / 2;
-  constexpr int RS = KS * KS;
-  constexpr int CH = 64;                   // m per chunk (2 halves of 32)
-  constexpr int WR = (BK >= 64) ? 2 : 1;   // wave rows (k dim)
-  constexpr int WC = 8 / WR;               // wave cols (rsc dim)
-  constexpr int FK32 = BK / WR / 32;       // 32-wide k frags per wave
-  constexpr int FR32 = BR / WC / 32;       // 32-wide rsc frags per wave
-  static_assert(FK32 >= 1 && FR32 >= 1, "tile too small for 8 waves");
-  constexpr int SSA = (BK / 16) * KBS_H;   // A elems per ring slot
-  constexpr int SSB = (BR / 16) * KBS_H;
-  constexpr int AUC = BK * 4;              // useful A chunks per half
-  constexpr int SB = BR / 128;             // B slots per thread (exact)
-  constexpr int PIECES = 1 + SB;           // glds instr per wave per half
-  const int KG = RS * Cp;
-  const long M = (long)N * H * W;
-  const int HW = H * W;
-
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  bf16_t* lA = reinterpret_cast<bf16_t*>(smem);  // 4 x SSA
-  bf16_t* lB = lA + 4 * SSA;                     // 4 x SSB
-
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wid = tid >> 6;
-  const int wr = wid / WC;
-  const int wc = wid % WC;
-  const int kt0 = blockIdx.x * BK;
-  const int rt0 = blockIdx.y * BR;
-
-  // ---- A slot (dY): strided over threads so every WAVE issues exactly
-  //      one (possibly exec-masked) glds — uniform vmcnt counts ----
-  constexpr int ASTR = 512 / AUC > 0 ? (AUC >= 512 ? 1 : 512 / AUC) : 1;
-  const bool aAct = (tid % ASTR) == 0;
-  const int aU = tid / ASTR;
-  const int aKblk = aU >> 6;
-  const int aMblk = (aU >> 3) & 7;
-  const int aR = aU & 7;
-  const int aMoff = aMblk * 4 + (aR >> 1);          // m within half
-  const int aCol = kt0 + aKblk * 16 + (aR & 1) * 8; // dY channel
-  const int aDst = aKblk * KBS_H + aMblk * TR_MBS + aR * 8;
-
-  // ---- B slots (X gather) ----
-  // NOTE fixed [3] bound (SB <= 3 for every instantiation): sizing these
-  // with the template-dependent SB makes this toolchain's HOST pass
-  // silently fail to emit the kernel's __device_stub__ when the array is
-  // captured by the stage lambda (undefined symbol at dlopen).
-  static_assert(SB <= 3, "bump the fixed B-slot array bound");
-  int bMoff[3], bC_[3], bDy[3], bDx[3], bDst[3];
-  bool bPad[3];
-#pragma unroll
-  for (int s = 0; s < SB; ++s) {
-    const int u = tid + s * 512;
-    const int kblk = u >> 6;
-    const int mblk = (u >> 3) & 7;
-    const int r = u & 7;
-    const int rsc = rt0 + kblk * 16 + (r & 1) * 8;
-    bPad[s] = rsc >= KG;
-    bMoff[s] = mblk * 4 + (r >> 1);
-    const int tap = rsc >> log2Cp;
-    bC_[s] = rsc & (Cp - 1);
-    bDy[s] = tap / KS;
-    bDx[s] = tap - bDy[s] * KS;
-    bDst[s] = kblk * KBS_H + mblk * TR_MBS + r * 8;
-  }
-
-  const long nChunks = (M + CH - 1) / CH;
-  const long myChunks =
-      (nChunks - blockIdx.z + splitm - 1) / splitm;  // z < nChunks always
-  const long nHalf = 2 * myChunks;
-
-  auto stage = [&](long j) {
-    const int slot = (int)(j & 3);
-    const long mbase =
-        ((long)blockIdx.z + (j >> 1) * splitm) * CH + (j & 1) * 32;
-    {
-      // every lane of every wave issues exactly ONE A glds (uniform vmcnt
-      // counts); lanes with no useful slot (AUC < 512) write the kblk-0
-      // pad region, which no tr read ever touches.
-      const bf16_t* src = Zero16w;
-      const long m = mbase + aMoff;
-      int dst = 8 * TR_MBS;
-      if (aAct) {
-        dst = aDst;
-        if (m < M) src = dY + m * Kp + aCol;
-      }
-      __builtin_amdgcn_global_load_lds(src, lA + slot * SSA + dst, 16, 0, 0);
-    }
-#pragma unroll
-    for (int s = 0; s < SB; ++s) {
-      const bf16_t* src = Zero16w;
-      const long m = mbase + bMoff[s];
-      if (!bPad[s] && m < M) {
-        unsigned n = magic_div((unsigned)m, mulHW);
-        unsigned rem = (unsigned)m - n * (unsigned)HW;
-        unsigned oy = magic_div(rem, mulW);
-        int ox = (int)(rem - oy * (unsigned)W);
-        int iy = (int)oy + bDy[s] - PAD, ix = ox + bDx[s] - PAD;
-        if (iy >= 0 && iy < H && ix >= 0 && ix < W)
-          src = X + (((long)((int)n * H + iy) * W + ix) << log2Cp) + bC_[s];
-      }
-      __builtin_amdgcn_global_load_lds(src, lB + slot * SSB + bDst[s], 16, 0,
-                                       0);
-    }
-  };
-
-  f32x16 acc32[FK32][FR32];
-#pragma unroll
-  for (int a = 0; a < FK32; ++a)
-#pragma unroll
-    for (int b = 0; b < FR32; ++b)
-#pragma unroll
-      for (int e = 0; e < 16; ++e) acc32[a][b][e] = 0.f;
-
-  // per-lane tr-read byte offsets (32x32x16 fragment): lane l -> col
-  // (l&31) => col-block (l>>4)&1, m = (l>>5)*8 + e => mblk (l>>5)*2 (+1)
-  const unsigned trL =
-      (unsigned)((((lane >> 4) & 1) * KBS_H + (lane >> 5) * 2 * TR_MBS +
-                  (lane & 15) * 4) * 2);
-  const unsigned aBase = (unsigned)(unsigned long long)lA + trL;
-  const unsigned bBase = (unsigned)(unsigned long long)lB + trL;
-
-  // counted vmcnt: wait until at most `halves` halves' pieces remain
-  auto wait_halves = [&](int halves) {
-    if constexpr (PIECES == 4) {
-      if (halves == 2) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-      else if (halves == 1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    } else if constexpr (PIECES == 3) {
-      if (halves == 2) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-      else if (halves == 1) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
-      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    } else {
-      if (halves == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      else if (halves == 1) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-  };
-
-  for (long j = 0; j < nHalf && j < 3; ++j) stage(j);
-
-  for (long j = 0; j < nHalf; ++j) {
-    // own pieces for half j landed (j+1, j+2 may stay in flight — T4)
-    {
-      const long r_ = nHalf - 1 - j;  // no min(long,long): host pass lacks
-      wait_halves(r_ > 2 ? 2 : (int)r_);  // the overload (silent stub kill)
-    }
-    __builtin_amdgcn_s_barrier();
-    if (j + 3 < nHalf) stage(j + 3);  // into slot (j-1)&3: reads done by all
-                                      // waves before they arrived above
-    const unsigned aOff = (unsigned)((j & 3) * SSA) * 2;
-    const unsigned bOff = (unsigned)((j & 3) * SSB) * 2;
-    constexpr int RPK = 2 * (FK32 + FR32);  // tr reads per kstep
-    bf16x4 aT[2][FK32][2], bT[2][FR32][2];  // 2-deep kstep ring
-    auto rdk = [&](int kst, int ring) {
-#pragma unroll
-      for (int f = 0; f < FK32; ++f) {
-        const unsigned base =
-            aBase + aOff +
-            (unsigned)(((wr * (BK / WR / 16) + f * 2) * KBS_H +
-                        kst * 4 * TR_MBS) * 2);
-        aT[ring][f][0] = ds_tr16(base);
-        aT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
-      }
-#pragma unroll
-      for (int f = 0; f < FR32; ++f) {
-        const unsigned base =
-            bBase + bOff +
-            (unsigned)(((wc * (BR / WC / 16) + f * 2) * KBS_H +
-                        kst * 4 * TR_MBS) * 2);
-        bT[ring][f][0] = ds_tr16(base);
-        bT[ring][f][1] = ds_tr16(base + TR_MBS * 2);
-      }
-    };
-    rdk(0, 0);
-#pragma unroll
-    for (int kst = 0; kst < 2; ++kst) {
-      const int ring = kst & 1;
-      if (kst == 0) {
-        rdk(1, 1);
-        // wait for kstep 0's reads only; kstep 1's RPK stay in flight
-        if constexpr (RPK == 10)
-          asm volatile("s_waitcnt lgkmcnt(10)" ::: "memory");
-        else if constexpr (RPK == 8)
-          asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
-        else if constexpr (RPK == 6)
-          asm volatile("s_waitcnt lgkmcnt(6)" ::: "memory");
-        else
-          asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      }
-      __builtin_amdgcn_sched_barrier(0);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int fa = 0; fa < FK32; ++fa)
-#pragma unroll
-        for (int fb = 0; fb < FR32; ++fb) {
-          const bf16x8 av = __builtin_shufflevector(
-              aT[ring][fa][0], aT[ring][fa][1], 0, 1, 2, 3, 4, 5, 6, 7);
-          const bf16x8 bv = __builtin_shufflevector(
-              bT[ring][fb][0], bT[ring][fb][1], 0, 1, 2, 3, 4, 5, 6, 7);
-          acc32[fa][fb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              av, bv, acc32[fa][fb], 0, 0, 0);
-        }
-      __builtin_amdgcn_s_setprio(0);
-    }
-  }
-
-  // ---- 32x32 epilogue: D col = lane&31 (rsc), row = (r&3) + 8*(r>>2)
-  //      + 4*(lane>>5) (k); scatter-add into NCHW fp32 dW ----
-#pragma unroll
-  for (int fa = 0; fa < FK32; ++fa) {
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-      const int k = kt0 + wr * (BK / WR) + fa * 32 + row;
-      if (k >= K) continue;
-#pragma unroll
-      for (int fb = 0; fb < FR32; ++fb) {
-        const int rsc = rt0 + wc * (BR / WC) + fb * 32 + (lane & 31);
-        if (rsc >= KG) continue;
-        const int tap = rsc >> log2Cp;
-        const int c = rsc & (Cp - 1);
-        if (c >= C) continue;
-        const int dy_ = tap / KS, dx_ = tap - (tap / KS) * KS;
-        atomicAdd(&dW[(((long)k * C + c) * KS + dy_) * KS + dx_],
-                  acc32[fa][fb][r]);
-      }
-    }
-  }
-}
-
-template __global__ void k_conv_wgrad8<7, 128, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<5, 128, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<3, 128, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<7, 64, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<5, 64, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<3, 64, 384>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<1, 64, 128>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<7, 32, 256>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-template __global__ void k_conv_wgrad8<5, 32, 256>(
-    const bf16_t*, const bf16_t*, float*, int, int, int, int, int,
-    int, int, int, int, unsigned long long, unsigned long long,
-    const bf16_t*);
-
-// ---------------------------------------------------------------------------
 // Small-K weight gradient (K <= 4: the 64->3 / 32->3 output convs).
 // The MFMA path wastes 13/16 rows of every fragment on Kp padding there;
 // a VALU outer-product reduction is ~10x cheaper. Each block owns one
@@ -1794,81 +1492,6 @@ void conv2d_wgrad(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
   }
   const int KG = (int)(ks * ks) * Cp;
   const int BK = std::min(Kp, 128);
-  // ---- v4: 8-wave counted-vmcnt phase pipeline for the WaterNet layer
-  //      shapes (see k_conv_wgrad8). WN_WGRAD_V4=0 falls back to v3. ----
-  static const bool use_v4 = [] {
-    const char* e = getenv("WN_WGRAD_V4");
-    return e == nullptr || atoi(e) != 0;
-  }();
-  if (use_v4) {
-    static thread_local at::Tensor zero16v4;
-    if (!zero16v4.defined() || zero16v4.device() != x.device())
-      zero16v4 = at::zeros({8}, x.options());
-    hipStream_t s4 = at::cuda::getCurrentHIPStream();
-    const long M4 = (long)N * H * W;
-    const long nCh4 = (M4 + 63) / 64;
-    auto l4 = [&](auto ksc, auto bkc, auto brc) -> bool {
-      constexpr int KSV = decltype(ksc)::value;
-      constexpr int BKV = decltype(bkc)::value;
-      constexpr int BRV = decltype(brc)::value;
-      const int gx4 = (Kp + BKV - 1) / BKV;
-      const int gy4 = (KG + BRV - 1) / BRV;
-      int split4 = std::max(1, 512 / std::max(1, gx4 * gy4));
-      split4 = (int)std::min<long>(split4, nCh4);
-      static const int split_cap = [] {  // debug: WN_WGRAD_SPLIT caps z
-        const char* e = getenv("WN_WGRAD_SPLIT");
-        return e ? atoi(e) : 0;
-      }();
-      if (split_cap > 0) split4 = std::min(split4, split_cap);
-      const size_t lds4 =
-          (size_t)4 * ((BKV / 16 + BRV / 16) * (8 * TR_MBS + 16)) *
-          sizeof(bf16_t);
-      hipLaunchKernelGGL((k_conv_wgrad8<KSV, BKV, BRV>),
-                         dim3(gx4, gy4, split4), dim3(512), lds4, s4,
-                         (const bf16_t*)dy.data_ptr(),
-                         (const bf16_t*)x.data_ptr(), dw.data_ptr<float>(),
-                         N, H, W, Cp, log2i(Cp), Kp, K, C, split4,
-                         MagicDiv::make((unsigned)(H * W)).mul,
-                         MagicDiv::make((unsigned)W).mul,
-                         (const bf16_t*)zero16v4.data_ptr());
-      HIP_CHECK_LAST();
-      return true;
-    };
-    bool done4 = false;
-    using std::integral_constant;
-    if (BK == 128) {
-      if (ks == 7) done4 = l4(integral_constant<int, 7>{},
-                              integral_constant<int, 128>{},
-                              integral_constant<int, 384>{});
-      else if (ks == 5) done4 = l4(integral_constant<int, 5>{},
-                                   integral_constant<int, 128>{},
-                                   integral_constant<int, 384>{});
-      else if (ks == 3) done4 = l4(integral_constant<int, 3>{},
-                                   integral_constant<int, 128>{},
-                                   integral_constant<int, 384>{});
-    } else if (BK == 64) {
-      if (ks == 7) done4 = l4(integral_constant<int, 7>{},
-                              integral_constant<int, 64>{},
-                              integral_constant<int, 384>{});
-      else if (ks == 5) done4 = l4(integral_constant<int, 5>{},
-                                   integral_constant<int, 64>{},
-                                   integral_constant<int, 384>{});
-      else if (ks == 3) done4 = l4(integral_constant<int, 3>{},
-                                   integral_constant<int, 64>{},
-                                   integral_constant<int, 384>{});
-      else if (ks == 1) done4 = l4(integral_constant<int, 1>{},
-                                   integral_constant<int, 64>{},
-                                   integral_constant<int, 128>{});
-    } else if (BK == 32) {
-      if (ks == 7) done4 = l4(integral_constant<int, 7>{},
-                              integral_constant<int, 32>{},
-                              integral_constant<int, 256>{});
-      else if (ks == 5) done4 = l4(integral_constant<int, 5>{},
-                                   integral_constant<int, 32>{},
-                                   integral_constant<int, 256>{});
-    }
-    if (done4) return;
-  }
   constexpr int WG_CH = 64;  // m rows per chunk (glds double-buffer budget)
   const int gx = (Kp + BK - 1) / BK, gy = (KG + 127) / 128;
   // split so gx*gy*split fills 256 CUs x ~3 resident blocks
