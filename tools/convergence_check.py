@@ -30,8 +30,23 @@ from waternet_amd.utils.metrics import (
 )
 
 DEV = "cuda:0"
-STEPS = 150
-BS, H, W = 8, 64, 64
+
+import argparse
+
+_ap = argparse.ArgumentParser()
+_ap.add_argument("--steps", type=int, default=150)
+_ap.add_argument("--bs", type=int, default=8)
+_ap.add_argument("--hw", type=int, default=64)
+_ap.add_argument("--flagship", action="store_true",
+                 help="BASELINE config 2 shape: bs=16, 112x112, 400 steps")
+_ap.add_argument("--out", default="convergence.json")
+_args = _ap.parse_args()
+if _args.flagship:
+    _args.steps, _args.bs, _args.hw = 400, 16, 112
+    if _args.out == "convergence.json":
+        _args.out = "convergence_flagship.json"
+STEPS = _args.steps
+BS, H, W = _args.bs, _args.hw, _args.hw
 
 
 def make_data():
@@ -83,8 +98,10 @@ def run(eager: bool):
 
 native = run(eager=False)
 eager = run(eager=True)
-result = {"steps": STEPS, "native_bf16": native, "eager_fp32": eager}
-outp = pathlib.Path(__file__).resolve().parent.parent / "profiles" / "convergence.json"
+result = {"steps": STEPS, "bs": BS, "hw": H,
+          "native_bf16": native, "eager_fp32": eager}
+outp = (pathlib.Path(__file__).resolve().parent.parent / "profiles"
+        / _args.out)
 outp.write_text(json.dumps(result, indent=1))
 nf, ef = native[-1], eager[-1]
 print("final native:", nf)
