@@ -85,14 +85,28 @@ def run(eager: bool):
         loss.backward()
         opt.step()
         if step % 10 == 0 or step == STEPS - 1:
+            # evaluate on the FULL fixed pool (not the current train batch)
+            # so checkpoints are comparable across engines/steps
             with torch.no_grad():
-                ssim = structural_similarity_index_measure(
-                    preds=out.detach(), target=ref_f).item()
-                psnr = peak_signal_noise_ratio(
-                    preds=out.detach(), target=ref_f, data_range=1.0).item()
+                ssims, psnrs, mses = [], [], []
+                for rb, fb in zip(raws, refs):
+                    rt = torch.from_numpy(rb).to(DEV)
+                    ft = torch.from_numpy(fb).to(DEV)
+                    wb2, gc2, he2 = gpu_transform_batch(rt)
+                    r2, w2, g2, h2 = (e.u8_to_nchw(t)
+                                      for t in (rt, wb2, gc2, he2))
+                    f2 = e.u8_to_nchw(ft)
+                    o2 = model(r2, w2, h2, g2)
+                    ssims.append(structural_similarity_index_measure(
+                        preds=o2, target=f2).item())
+                    psnrs.append(peak_signal_noise_ratio(
+                        preds=o2, target=f2, data_range=1.0).item())
+                    mses.append(torch.mean(
+                        (255.0 * (o2 - f2)) ** 2).item())
             curve.append({"step": step, "loss": loss.item(),
-                          "mse255": mloss.item(), "ssim": ssim,
-                          "psnr": psnr})
+                          "mse255": float(np.mean(mses)),
+                          "ssim": float(np.mean(ssims)),
+                          "psnr": float(np.mean(psnrs))})
     return curve
 
 

@@ -29,10 +29,10 @@ class ClockSampler(threading.Thread):
         super().__init__(daemon=True)
         self.period = period
         self.samples = []
-        self._stop = threading.Event()
+        self._halt = threading.Event()
 
     def run(self):
-        while not self._stop.is_set():
+        while not self._halt.is_set():
             try:
                 out = subprocess.run(
                     ["rocm-smi", "--showgpuclocks", "--showpower",
@@ -53,7 +53,7 @@ class ClockSampler(threading.Thread):
             time.sleep(self.period)
 
     def stop(self):
-        self._stop.set()
+        self._halt.set()
 
 
 def main():

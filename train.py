@@ -117,9 +117,8 @@ def _clean_state_dict(model):
 
 def _train_epoch_fast(engine, loader, batch_size, device, epoch_num,
                       total_epochs, progress):
-    from waternet_amd.engine.fast import FastStepEngine  # noqa: F401
-
     snap = engine.metrics_snapshot()
+    t0 = timer()
     iterator = loader
     if progress:
         try:
@@ -130,15 +129,24 @@ def _train_epoch_fast(engine, loader, batch_size, device, epoch_num,
                             bar_format="{l_bar}{bar:20}{r_bar}")
         except ImportError:
             pass
+    n_images = 0
     for batch in iterator:
         raw, ref = batch["raw"], batch["ref"]
+        n_images += raw.size(0)
         if raw.size(0) == batch_size:
             engine.load_batch(raw, ref)
             engine.step()
         else:  # ragged tail batch: eager step at its true size
             engine.step_batch(raw.to(device, non_blocking=True),
                               ref.to(device, non_blocking=True))
-    m = engine.metrics_since(snap)  # ONE host sync per epoch
+    m = engine.metrics_since(snap)  # ONE host sync per epoch (joins the GPU)
+    wall = timer() - t0
+    if progress and n_images:
+        # wall includes queueing + the GPU completing every queued step
+        # (metrics_since syncs); this is the per-rank TRAIN throughput —
+        # the number to compare against bench.py
+        print(f"    [fast] epoch train wall {wall:.3f}s = "
+              f"{n_images / wall:.0f} img/s")
     return {
         "mse": m["mse255"], "ssim": m["ssim"], "psnr": m["psnr"],
         "perceptual_loss": m["perceptual"], "loss": m["loss"],
