@@ -37,21 +37,6 @@
 #define WN_MFMA_KMAP 0
 #endif
 
-// T1 XCD-contiguous block swizzle (guide §6 G16 / T1): dispatch round-robins
-// blocks across the 8 XCDs (block b -> XCD b%8), so CONSECUTIVE tiles — which
-// share conv halo rows — land on different (non-coherent) L2s and re-fetch
-// from L3/HBM. At 512^2 the activations exceed the 256 MB L3, making the
-// k5/k3 tap re-reads HBM traffic (PMC r04: igemm8 MFMA duty 7-8% with 52-56%
-// SQ_WAIT_ANY). Remapping so each XCD owns a contiguous tile range keeps a
-// tile's 3-5 halo rows resident in its own 4 MB L2. Bijective for any nwg
-// (the naive (b%8)*ceil+b/8 remap is not when nwg%8 != 0).
-WN_DEVFN int xcd_swz(int bid, int nwg) {
-  if (nwg < 64) return bid;  // too few tiles for locality to matter
-  const int q = nwg >> 3, r = nwg & 7;
-  const int xcd = bid & 7, i = bid >> 3;
-  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + i;
-}
-
 // ---------------------------------------------------------------------------
 // Forward / dgrad implicit GEMM
 // ---------------------------------------------------------------------------
@@ -96,10 +81,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_igemm(
   const int wid = tid >> 6;
   const int wr = wid / WN;
   const int wc = wid % WN;
-  // SPLITK shapes are tiny-M (no halo-reuse win, and z slices K): identity
-  const long m0 =
-      (long)(SPLITK ? (int)blockIdx.x
-                    : xcd_swz(blockIdx.x, gridDim.x)) * BM;
+  const long m0 = (long)blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
 
   // ---- per-thread A-slot geometry (2 slots) ----
@@ -366,7 +348,7 @@ __global__ __launch_bounds__(512, 1) void k_conv_igemm8(
   const int wid = tid >> 6;
   const int wr = wid >> 1;
   const int wc = wid & 1;
-  const long m0 = (long)xcd_swz(blockIdx.x, gridDim.x) * BM;
+  const long m0 = (long)blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const int HW = H * W;
 
@@ -750,13 +732,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const int wr = wid / WC;
   const int wc = wid % WC;
   const int kt0 = blockIdx.x * BK;
-  // y-fastest (y,z) flat remap: each XCD owns contiguous flats = complete
-  // rsc-tile runs of one split index -> the blocks sharing an m-chunk's
-  // staged dY / halo-overlapped X live on one L2
-  const int flat_ =
-      xcd_swz(blockIdx.y + gridDim.y * blockIdx.z, gridDim.y * gridDim.z);
-  const int bz_ = flat_ / gridDim.y;
-  const int rt0 = (flat_ - bz_ * gridDim.y) * BR;
+  const int rt0 = blockIdx.y * BR;
 
   // ---- static slot geometry (sigma -> image position -> source) ----
   int aM_[SA], aCol[SA];
@@ -849,7 +825,7 @@ __global__ __launch_bounds__(256, 2) void k_conv_wgrad(
   const unsigned aTrM = (unsigned)(unsigned long long)lA + trM32;
   const unsigned bTrM = (unsigned)(unsigned long long)lB + trM32;
 
-  long chunk = bz_;
+  long chunk = blockIdx.z;
   int buf = 0;
   if (chunk < nChunks) {
     stage(0, chunk * CH);
