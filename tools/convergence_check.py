@@ -112,18 +112,41 @@ def run(eager: bool):
 
 native = run(eager=False)
 eager = run(eager=True)
-result = {"steps": STEPS, "bs": BS, "hw": H,
+
+# Trajectory comparison: single end-points oscillate +-15% at lr=1e-3 on a
+# 4-batch pool, so compare TAIL MEANS (last 5 checkpoints) and the mean
+# relative loss gap over the back half of training.
+import statistics
+
+
+def tail_mean(curve, key, n=5):
+    return statistics.mean(x[key] for x in curve[-n:])
+
+
+half = len(native) // 2
+back_rel = statistics.mean(
+    abs(n_["loss"] - e_["loss"]) / max(abs(e_["loss"]), 1e-9)
+    for n_, e_ in zip(native[half:], eager[half:]))
+summary = {
+    "native_tail": {k: tail_mean(native, k)
+                    for k in ("loss", "mse255", "ssim", "psnr")},
+    "eager_tail": {k: tail_mean(eager, k)
+                   for k in ("loss", "mse255", "ssim", "psnr")},
+    "back_half_mean_rel_loss_gap": back_rel,
+}
+result = {"steps": STEPS, "bs": BS, "hw": H, "summary": summary,
           "native_bf16": native, "eager_fp32": eager}
 outp = (pathlib.Path(__file__).resolve().parent.parent / "profiles"
         / _args.out)
 outp.write_text(json.dumps(result, indent=1))
-nf, ef = native[-1], eager[-1]
-print("final native:", nf)
-print("final eager :", ef)
-rel = abs(nf["loss"] - ef["loss"]) / (abs(ef["loss"]) + 1e-9)
-print(f"final loss rel diff: {rel:.3f}")
-assert nf["loss"] < native[0]["loss"] * 0.7, "native loss did not decrease"
-assert ef["loss"] < eager[0]["loss"] * 0.7, "eager loss did not decrease"
-assert rel < 0.15, "native and eager trajectories diverged"
-assert abs(nf["psnr"] - ef["psnr"]) < 1.5
+nt, et = summary["native_tail"], summary["eager_tail"]
+print("tail native:", nt)
+print("tail eager :", et)
+rel = abs(nt["loss"] - et["loss"]) / (abs(et["loss"]) + 1e-9)
+print(f"tail loss rel diff: {rel:.3f}; "
+      f"back-half mean rel loss gap: {back_rel:.3f}")
+assert native[-1]["loss"] < native[0]["loss"] * 0.7, "native did not train"
+assert eager[-1]["loss"] < eager[0]["loss"] * 0.7, "eager did not train"
+assert rel < 0.15, "native and eager tail losses diverged"
+assert abs(nt["psnr"] - et["psnr"]) < 1.0, "tail PSNR gap too large"
 print("CONVERGENCE PARITY OK")
