@@ -184,3 +184,63 @@ def test_ddp_loss_parity_vs_single():
     for r, w in results.items():
         assert torch.allclose(w, w_single, atol=1e-5), (
             r, (w - w_single).abs().max())
+
+
+def _run_arena_ddp_worker(rank, world_size, port, q):
+    """bench.py's exact DDP arithmetic on CPU/gloo: FusedAdam flat arena,
+    rank-0 master broadcast, grads.div_(world) + SUM all-reduce before
+    opt.step() — the path the driver's 8-GPU scale run exercises."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    try:
+        from waternet_amd.models.waternet import WaterNet
+        from waternet_amd.ops.adam import FusedAdam
+        from waternet_amd.parallel import init_distributed
+
+        init_distributed(backend="gloo")
+        torch.manual_seed(1234 + rank)  # per-rank init, as bench.py
+        model = WaterNet()
+        opt = FusedAdam(model.parameters(), lr=1e-3, model=model)
+        dist.broadcast(opt.master, src=0)  # bench.py:128
+
+        torch.manual_seed(500 + rank)  # per-rank data shard
+        x = torch.rand(2, 3, 16, 16)
+        ref = torch.rand(2, 3, 16, 16)
+        for _ in range(2):
+            out = model(x, x, x, x)
+            loss = torch.mean((255.0 * (out - ref)) ** 2)
+            opt.zero_grad()
+            loss.backward()
+            opt.grads.div_(world_size)  # fast.py DDP branch
+            dist.all_reduce(opt.grads)
+            opt.step()
+        q.put((rank, opt.master.double().sum().item(),
+               model.cmg.conv1.weight.detach().numpy().tolist()))
+        dist.destroy_process_group()
+    except Exception as e:  # noqa: BLE001
+        q.put((rank, "ERROR", repr(e)))
+
+
+def test_bench_arena_ddp_flow_world2():
+    world_size = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [
+        ctx.Process(target=_run_arena_ddp_worker,
+                    args=(r, world_size, 29817, q))
+        for r in range(world_size)
+    ]
+    for p in procs:
+        p.start()
+    results = [q.get() for _ in range(world_size)]
+    for p in procs:
+        p.join(timeout=120)
+    errors = [r for r in results if r[1] == "ERROR"]
+    assert not errors, errors
+    # after broadcast + identical averaged grads, ranks stay bit-identical
+    sums = {round(r[1], 9) for r in results}
+    assert len(sums) == 1, results
+    w = {r[0]: torch.tensor(r[2]) for r in results}
+    assert torch.equal(w[0], w[1])
