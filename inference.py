@@ -107,12 +107,16 @@ def enhance_frame(model, rgb: np.ndarray, device) -> np.ndarray:
     pipeline (preprocess + forward + postprocess) runs as a hipGraph-
     captured on-device pipeline; on CPU the reference-semantics numpy
     transforms + eager model run."""
-    if device.type == "cuda" and rgb.shape[0] % 8 == 0 and rgb.shape[1] % 8 == 0:
+    if device.type == "cuda":
         from waternet_amd.ops import native_available
 
         if native_available():
-            eng = _gpu_engine(model, rgb.shape[0], rgb.shape[1], device)
-            return eng.infer_frame(rgb)
+            from waternet_amd.engine.inferencer import pad8
+
+            padded, h, w = pad8(rgb)
+            eng = _gpu_engine(model, padded.shape[0], padded.shape[1],
+                              device)
+            return eng.infer_frame(padded)[:h, :w]
     wb, gc, he = transform(rgb)
     rgb_ten = arr2ten(rgb, add_batch_dim=True).to(device)
     wb_ten = arr2ten(wb, add_batch_dim=True).to(device)

@@ -43,19 +43,24 @@ class Server:
         if h > MAX_DIM or w > MAX_DIM:
             raise ValueError(
                 f"image {w}x{h} exceeds the {MAX_DIM}px per-side limit")
-        if self.device.type == "cuda" and h % 8 == 0 and w % 8 == 0:
-            eng = self._engines.get((h, w))
-            if eng is None:
-                from waternet_amd.engine.inferencer import InferenceEngine
+        if self.device.type == "cuda":
+            from waternet_amd.engine.inferencer import InferenceEngine, pad8
 
-                eng = InferenceEngine(self.model, h, w, device=self.device)
+            # any resolution runs on GPU: reflect-pad to /8 (the CLAHE
+            # tile grid), crop back after — also coalesces the engine
+            # cache onto /8 keys
+            padded, oh, ow = pad8(rgb_u8)
+            ph, pw = padded.shape[:2]
+            eng = self._engines.get((ph, pw))
+            if eng is None:
+                eng = InferenceEngine(self.model, ph, pw, device=self.device)
                 while len(self._engines) >= MAX_ENGINES:
                     self._engines.popitem(last=False)
-                self._engines[(h, w)] = eng
+                self._engines[(ph, pw)] = eng
             else:
-                self._engines.move_to_end((h, w))
-            return eng.infer_frame(rgb_u8)
-        # CPU / odd-size fallback: reference transforms + eager forward
+                self._engines.move_to_end((ph, pw))
+            return eng.infer_frame(padded)[:oh, :ow]
+        # CPU fallback: reference transforms + eager forward
         from waternet_amd.data.bridge import arr2ten, ten2arr
         from waternet_amd.data.transforms import transform
 

@@ -340,10 +340,12 @@ def test_serve_enhance_gpu():
     assert (64, 96) in srv._engines
     out2 = srv.enhance(img)  # cached-engine path
     assert np.array_equal(out, out2)
-    # odd size falls back to the eager path
+    # odd sizes run the GPU engine via reflect-pad + crop (cache keyed
+    # on the padded /8 size)
     img2 = rng.integers(0, 256, size=(33, 45, 3), dtype=np.uint8)
     out3 = srv.enhance(img2)
     assert out3.shape == (33, 45, 3)
+    assert (40, 48) in srv._engines
     # dimension cap enforced
     import serve as serve_mod
 

@@ -19,6 +19,22 @@ from waternet_amd.ops import ext
 from waternet_amd.ops.preprocess import gpu_transform_batch
 
 
+def pad8(rgb_u8: np.ndarray):
+    """Reflect-pad an HWC frame to the next multiple of 8 per side (the
+    GPU CLAHE tile-grid requirement). Returns (padded, orig_h, orig_w).
+
+    Note: the padded rows/cols (<= 7 px, reflected) participate in the
+    WB quantile and CLAHE tile statistics — a boundary-only deviation
+    from the CPU transforms, accepted so ANY resolution runs the
+    hipGraph GPU pipeline (crop back with [:h, :w] after)."""
+    h, w = rgb_u8.shape[:2]
+    ph, pw = (-h) % 8, (-w) % 8
+    if ph == 0 and pw == 0:
+        return rgb_u8, h, w
+    return (np.pad(rgb_u8, ((0, ph), (0, pw), (0, 0)), mode="reflect"),
+            h, w)
+
+
 class InferenceEngine:
     def __init__(self, model, height, width, device="cuda:0",
                  use_graph=True):
