@@ -457,3 +457,69 @@ def test_ssim_nhwc_parity(ext):
     got = ssim_nhwc(to_nhwc_bf16(a, 16), to_nhwc_bf16(b, 16), 3, 1.0)
     want = _ssim_torch(q(a), q(b), 1.0, 11, 1.5, 0.01, 0.03)
     assert torch.allclose(got, want.to(got.dtype), atol=2e-3), (got, want)
+
+
+def test_conv_shape_fuzz(ext):
+    """Randomized shapes within the engine envelope (stride-1 same-pad
+    ks in {1,3,5,7}, arbitrary N/H/W) — guards against dispatch/tile
+    edge cases beyond the 9 fixed layer shapes."""
+    from waternet_amd.ops.conv import ConvSpec, conv_bias_act, pow2_channels
+    import torch.nn as nn
+
+    rng = np.random.default_rng(31)
+    for _ in range(6):
+        ks = int(rng.choice([1, 3, 5, 7]))
+        C = int(rng.choice([3, 6, 12, 24, 32, 64, 100, 128]))
+        K = int(rng.choice([3, 16, 32, 48, 64, 128]))
+        N = int(rng.integers(1, 5))
+        H = int(rng.integers(max(ks, 4), 40))
+        W = int(rng.integers(max(ks, 4), 40))
+        torch.manual_seed(int(rng.integers(0, 1000)))
+        mod = nn.Conv2d(C, K, ks, padding="same").to(DEV)
+        spec = ConvSpec(mod, act=1)
+        x = torch.rand(N, C, H, W, device=DEV)
+        x_nhwc = to_nhwc_bf16(x, pow2_channels(C)).requires_grad_(True)
+        y = conv_bias_act(x_nhwc, spec)
+        dy = torch.randn_like(y)
+        dy[..., K:] = 0
+        y.backward(dy)
+
+        xq = q(x).requires_grad_(True)
+        wq = q(mod.weight.detach()).requires_grad_(True)
+        bq = mod.bias.detach().clone().requires_grad_(True)
+        yref = F.relu(F.conv2d(xq, wq, bq, padding=ks // 2))
+        yref.backward(from_nhwc(dy, K))
+        lbl = f"ks{ks} C{C} K{K} N{N} {H}x{W}"
+        yerr = (from_nhwc(y.detach(), K) - yref.detach()).abs().max().item()
+        ys = yref.detach().abs().max().item() + 1e-6
+        assert yerr / ys < 3e-2, f"fwd {lbl}: {yerr / ys}"
+        for got, ref, name in [
+            (from_nhwc(x_nhwc.grad, C), xq.grad, "dx"),
+            (mod.weight.grad, wq.grad, "dw"),
+            (mod.bias.grad, bq.grad, "db"),
+        ]:
+            scale = ref.abs().max().item() + 1e-6
+            err = (got - ref).abs().max().item() / scale
+            assert err < 4e-2, f"{name} {lbl}: rel err {err}"
+
+
+def test_preprocess_shape_fuzz(ext):
+    """GPU preprocess at varied /8 geometries vs the CPU reference."""
+    from waternet_amd.data.transforms import transform as cpu_transform
+    from waternet_amd.ops.preprocess import gpu_transform_batch
+
+    rng = np.random.default_rng(17)
+    for (h, w, n) in [(8, 8, 1), (16, 88, 2), (104, 24, 3), (72, 72, 1)]:
+        raw = rng.integers(0, 256, size=(n, h, w, 3), dtype=np.uint8)
+        wb_g, gc_g, he_g = gpu_transform_batch(
+            torch.from_numpy(raw).to(DEV))
+        for i in range(n):
+            wb_c, gc_c, he_c = cpu_transform(raw[i])
+            for got, ref, name, tol in [
+                (wb_g[i], wb_c, "wb", 2),
+                (gc_g[i], gc_c, "gc", 1),
+                (he_g[i], he_c, "he", 8),
+            ]:
+                d = (got.cpu().numpy().astype(int) - ref.astype(int))
+                assert np.abs(d).max() <= tol, (
+                    f"{name} {h}x{w}[{i}]: max diff {np.abs(d).max()}")
