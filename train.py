@@ -276,8 +276,20 @@ def main(argv=None):
         sidecar = Path(args.resume) / "last-trainstate.pt"
         if sidecar.exists():
             state = torch.load(sidecar, map_location="cpu")
+            opt_state = state.get("optimizer", {})
+            # FusedAdam stores its moments under "wn_fused"; a sidecar
+            # written by the eager engine (torch.optim.Adam) has per-param
+            # states FusedAdam cannot consume — loading it would silently
+            # reset the moments, so say so and skip the optimizer.
+            cross_engine = use_fast != ("wn_fused" in opt_state)
             try:
-                optimizer.load_state_dict(state["optimizer"])
+                if cross_engine:
+                    if rank0:
+                        print("WARNING: sidecar optimizer state was written "
+                              "by the other engine (fast<->eager); optimizer "
+                              "moments reset, LR schedule/epoch restored")
+                else:
+                    optimizer.load_state_dict(state["optimizer"])
                 scheduler.load_state_dict(state["scheduler"])
                 start_epoch = state["epoch"] + 1
                 if rank0:
