@@ -176,12 +176,18 @@ def main(argv=None):
     print(f"Using device: {device}")
 
     source = Path(args.source)
+    if not source.exists():
+        raise SystemExit(f"{args.source} does not exist!")
     if source.is_dir():
+        # both images and videos, as the reference (inference.py:128-135)
         sources = sorted(
-            p for p in source.iterdir() if p.suffix.lower() in IM_SUFFIXES
+            p for p in source.iterdir()
+            if p.suffix.lower() in IM_SUFFIXES
+            or p.suffix.lower() in VID_SUFFIXES
         )
     else:
         sources = [source]
+    print(f"Total images/videos: {len(sources)}")
 
     model = load_model(args.weights, device)
     savedir = make_savedir(args.name)
