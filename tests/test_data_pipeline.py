@@ -146,3 +146,33 @@ def test_train_engine_fast_requires_gpu():
         train_cli.main(["--engine", "fast", "--epochs", "1",
                         "--synthetic", "4", "--batch-size", "2",
                         "--height", "32", "--width", "32"])
+
+
+def test_pad8_helper():
+    import numpy as np
+
+    from waternet_amd.engine.inferencer import pad8
+
+    rng = np.random.default_rng(0)
+    x = rng.integers(0, 256, size=(45, 67, 3), dtype=np.uint8)
+    p, h, w = pad8(x)
+    assert (h, w) == (45, 67)
+    assert p.shape == (48, 72, 3)
+    assert np.array_equal(p[:45, :67], x)
+    # reflect padding: row 45 mirrors row 43 (reflect excludes the edge)
+    assert np.array_equal(p[45], p[43])
+    # already-aligned input passes through untouched
+    y = rng.integers(0, 256, size=(64, 64, 3), dtype=np.uint8)
+    p2, h2, w2 = pad8(y)
+    assert p2 is y and (h2, w2) == (64, 64)
+
+
+def test_hubconf_pretrained_offline_error():
+    """pretrained=True (the reference default) must raise a CLEAR error
+    when the download cannot proceed, not a bare network traceback."""
+    import pytest
+
+    import hubconf
+
+    with pytest.raises(RuntimeError, match="checkpoint=|pretrained=False"):
+        hubconf.waternet(pretrained=True, device="cpu")
