@@ -523,3 +523,21 @@ def test_preprocess_shape_fuzz(ext):
                 d = (got.cpu().numpy().astype(int) - ref.astype(int))
                 assert np.abs(d).max() <= tol, (
                     f"{name} {h}x{w}[{i}]: max diff {np.abs(d).max()}")
+
+
+def test_act_bwd_bias_fused_parity(ext):
+    """Fused act-backward + bias column sums vs the separate kernels."""
+    torch.manual_seed(3)
+    for act, K, Kp, n, h, w in [(1, 128, 128, 4, 12, 12),
+                                (2, 3, 16, 2, 16, 16),
+                                (1, 64, 64, 3, 9, 7)]:
+        dy = torch.randn(n, h, w, Kp, device=DEV).bfloat16().contiguous()
+        y = torch.rand(n, h, w, Kp, device=DEV).bfloat16().contiguous()
+        db_f = torch.randn(K, device=DEV)  # pre-existing accumulation
+        db_r = db_f.clone()
+        dpre_f = ext.act_bwd_bias(dy, y, act, db_f)
+        dpre_r = ext.act_bwd(dy, y, act)
+        ext.bias_grad(dpre_r, db_r)
+        assert torch.equal(dpre_f, dpre_r)
+        assert torch.allclose(db_f, db_r, atol=1e-2, rtol=1e-4), (
+            (db_f - db_r).abs().max())

@@ -31,6 +31,8 @@ std::vector<at::Tensor> fusion_bwd(const at::Tensor& dout,
                                    const at::Tensor& rce,
                                    const at::Tensor& rgc);
 at::Tensor act_bwd(const at::Tensor& dy, const at::Tensor& y, int64_t act);
+at::Tensor act_bwd_bias(const at::Tensor& dy, const at::Tensor& y,
+                        int64_t act, at::Tensor& db);
 at::Tensor normalize_vgg_fwd(const at::Tensor& x, int64_t Cp);
 at::Tensor normalize_vgg_bwd(const at::Tensor& dy, int64_t C);
 at::Tensor sqdiff255_sum(const at::Tensor& a, const at::Tensor& b,
@@ -81,6 +83,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fusion_fwd", &fusion_fwd);
   m.def("fusion_bwd", &fusion_bwd);
   m.def("act_bwd", &act_bwd);
+  m.def("act_bwd_bias", &act_bwd_bias,
+        "activation backward fused with bias-grad column sums");
   m.def("normalize_vgg_fwd", &normalize_vgg_fwd);
   m.def("normalize_vgg_bwd", &normalize_vgg_bwd);
   m.def("sqdiff255_sum", &sqdiff255_sum);

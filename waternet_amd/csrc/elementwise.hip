@@ -271,6 +271,67 @@ __global__ void k_act_bwd(const bf16_t* __restrict__ dy,
   }
 }
 
+// act_bwd fused with the bias-gradient column sums (K25 epilogue fusion):
+// dpre = dY * act'(Y) while accumulating sum_m dpre[m][k] per block into a
+// partial slab — saves the separate bias_grad kernel's full re-read of
+// dpre (the WaterNet layers' bias grads; the frozen VGG keeps plain
+// act_bwd). Thread->channel-octet mapping is STABLE across grid-stride
+// iterations because (gridDim*blockDim*8) % Kp == 0 (host guarantees
+// blocks*2048 divisible by Kp; Kp is a power of two <= 128).
+__global__ void k_act_bwd_bias(const bf16_t* __restrict__ dy,
+                               const bf16_t* __restrict__ y,
+                               bf16_t* __restrict__ dpre,
+                               float* __restrict__ part,  // [grid][Kp]
+                               long n8, int Kp, int act) {
+  __shared__ float red[128];  // Kp <= 128
+  if (threadIdx.x < Kp) red[threadIdx.x] = 0.f;
+  __syncthreads();
+  float s[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) s[e] = 0.f;
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (long i = i0; i < n8; i += (long)gridDim.x * blockDim.x) {
+    const bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
+    const bf16x8 vv = *reinterpret_cast<const bf16x8*>(y + i * 8);
+    bf16x8 r;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float g = bf2f(gv[e]), v = bf2f(vv[e]);
+      const float d = (act == ACT_RELU) ? (v > 0.f ? g : 0.f)
+                                        : g * v * (1.f - v);
+      r[e] = f2bf(d);
+      s[e] += d;
+    }
+    *reinterpret_cast<bf16x8*>(dpre + i * 8) = r;
+  }
+  const int c0 = (int)((i0 * 8) % Kp);
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(&red[c0 + e], s[e]);
+  __syncthreads();
+  if (threadIdx.x < Kp) part[(long)blockIdx.x * Kp + threadIdx.x] =
+      red[threadIdx.x];
+}
+
+// slab reduce into db (ACCUMULATES — the arena view may already hold
+// autograd-accumulated values)
+__global__ void k_abb_reduce(const float* __restrict__ part,
+                             float* __restrict__ db, int Kp, int K,
+                             int nblk) {
+  const int k = blockIdx.x;
+  if (k >= K) return;
+  __shared__ float red[256];
+  float s = 0.f;
+  for (int r = threadIdx.x; r < nblk; r += 256)
+    s += part[(long)r * Kp + k];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) db[k] += red[0];
+}
+
 // ---------------------------------------------------------------------------
 // ImageNet normalize fused with NCHW fp32 -> NHWC bf16 (and backward)
 // (train.py:111-116)
@@ -514,6 +575,34 @@ at::Tensor u8_to_nhwc(const at::Tensor& x, int64_t Cp) {
                      (bf16_t*)y.data_ptr(), NHW, (int)Cp);
   HIP_CHECK_LAST();
   return y;
+}
+
+at::Tensor act_bwd_bias(const at::Tensor& dy, const at::Tensor& y,
+                        int64_t act, at::Tensor& db) {
+  TORCH_CHECK(dy.is_cuda() && dy.dtype() == at::kBFloat16 && dy.dim() == 4);
+  TORCH_CHECK(db.is_cuda() && db.dtype() == at::kFloat);
+  const int Kp = (int)dy.size(3);
+  TORCH_CHECK(Kp <= 128 && (Kp & (Kp - 1)) == 0, "Kp power of two <= 128");
+  const int K = (int)db.size(0);
+  auto dpre = at::empty_like(dy);
+  const long n = dy.numel();
+  TORCH_CHECK(n % 8 == 0);
+  const long n8 = n / 8;
+  // blocks*blockDim*8 must be divisible by Kp: 256*8 = 2048 is divisible
+  // by every Kp <= 128, so any block count works
+  const int blocks = (int)std::min<long>(512, (n8 + 255) / 256);
+  auto part = at::empty({blocks, (long)Kp},
+                        dy.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(k_act_bwd_bias, dim3(blocks), dim3(TPB), 0,
+                     cur_stream(), (const bf16_t*)dy.data_ptr(),
+                     (const bf16_t*)y.data_ptr(), (bf16_t*)dpre.data_ptr(),
+                     part.data_ptr<float>(), n8, Kp, (int)act);
+  HIP_CHECK_LAST();
+  hipLaunchKernelGGL(k_abb_reduce, dim3(K), dim3(256), 0, cur_stream(),
+                     part.data_ptr<float>(), db.data_ptr<float>(), Kp, K,
+                     blocks);
+  HIP_CHECK_LAST();
+  return dpre;
 }
 
 at::Tensor normalize_nhwc_fwd(const at::Tensor& x) {

@@ -90,9 +90,20 @@ class ConvBiasAct(torch.autograd.Function):
         x, y = ctx.saved_tensors
         spec = ctx.spec
         dy = dy.contiguous()
-        dpre = ext().act_bwd(dy, y, spec.act) if spec.act != ACT_NONE else dy
-
         need_x, need_w, need_b = ctx.needs_input_grad[:3]
+        db_done = False
+        bias_view = (getattr(spec.mod.bias, "_wn_grad_view", None)
+                     if spec.mod.bias is not None else None)
+        if (spec.act != ACT_NONE and need_b and bias_view is not None):
+            # fused: dpre = dy*act'(y) with the bias column sums
+            # accumulated in the same pass (saves a full dpre re-read)
+            dpre = ext().act_bwd_bias(dy, y, spec.act, bias_view)
+            db_done = True
+        elif spec.act != ACT_NONE:
+            dpre = ext().act_bwd(dy, y, spec.act)
+        else:
+            dpre = dy
+
         dx = dw_ret = db_ret = None
         if need_x:
             # dgrad: conv of dpre (channels Kp) with rotated/transposed
@@ -107,7 +118,7 @@ class ConvBiasAct(torch.autograd.Function):
                 dw_ret = torch.zeros_like(spec.mod.weight,
                                           memory_format=torch.contiguous_format)
                 ext().conv2d_wgrad(dpre, x, dw_ret, spec.ks)
-        if need_b and spec.mod.bias is not None:
+        if need_b and spec.mod.bias is not None and not db_done:
             view = getattr(spec.mod.bias, "_wn_grad_view", None)
             if view is not None:
                 ext().bias_grad(dpre, view)
