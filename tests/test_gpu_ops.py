@@ -537,7 +537,13 @@ def test_act_bwd_bias_fused_parity(ext):
         db_r = db_f.clone()
         dpre_f = ext.act_bwd_bias(dy, y, act, db_f)
         dpre_r = ext.act_bwd(dy, y, act)
-        ext.bias_grad(dpre_r, db_r)
         assert torch.equal(dpre_f, dpre_r)
-        assert torch.allclose(db_f, db_r, atol=1e-2, rtol=1e-4), (
-            (db_f - db_r).abs().max())
+        # the fused kernel sums the UNROUNDED fp32 derivatives (more
+        # accurate than bias_grad's sum of bf16-rounded dpre), so compare
+        # against an fp32 recompute, not against bias_grad
+        g, v = dy.float(), y.float()
+        d = torch.where(v > 0, g, torch.zeros_like(g)) if act == 1 \
+            else g * v * (1 - v)
+        expected = db_r + d[..., :K].sum(dim=(0, 1, 2))
+        assert torch.allclose(db_f, expected, atol=2e-3, rtol=1e-4), (
+            (db_f - expected).abs().max())
