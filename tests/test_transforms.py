@@ -158,3 +158,21 @@ def test_white_balance_grayscale_path():
     assert np.array_equal(gray, before), "input mutated"
     # stretch reaches (close to) the full range
     assert out.max() >= 250 and out.min() <= 5
+
+
+def test_wb_degenerate_channels():
+    """All-black / single-channel images make the saturation ratio inf/nan
+    (the reference crashes in np.quantile there, data.py:40); we fall back
+    to a plain min/max stretch for the degenerate channel."""
+    import numpy as np
+
+    from waternet_amd.data.transforms import transform, white_balance_transform
+
+    black = np.zeros((16, 16, 3), dtype=np.uint8)
+    wb, gc, he = transform(black)
+    assert wb.shape == black.shape and wb.dtype == np.uint8
+    one = np.zeros((16, 16, 3), dtype=np.uint8)
+    one[:, :, 1] = 200
+    wbo = white_balance_transform(one)
+    assert wbo.dtype == np.uint8
+    assert wbo[:, :, 1].max() <= 255

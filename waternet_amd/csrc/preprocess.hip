@@ -91,6 +91,9 @@ __global__ void k_wb_params(const unsigned int* __restrict__ hist,
     double maxsum = fmax(sums[0], fmax(sums[1], sums[2]));
     double ratio = maxsum / sums[ch];
     double sat = 0.005 * ratio;
+    // degenerate channel (zero sum -> inf/nan sat, where the reference
+    // crashes): fall back to the un-saturated min/max stretch
+    if (!(isfinite(sat) && sat <= 0.5)) sat = 0.0;
     double lo = quantile_from_hist(h, HW, sat);
     double hi = quantile_from_hist(h, HW, 1.0 - sat);
     params[(n * 3 + ch) * 2 + 0] = (float)lo;

@@ -57,6 +57,12 @@ def white_balance_transform(im_rgb: np.ndarray) -> np.ndarray:
     out = np.empty_like(flat)
     for ch in range(p):
         lo_q, hi_q = sat_lo[ch], 1.0 - sat_hi[ch]
+        # Degenerate channels (a zero channel sum makes the sat ratio
+        # inf/nan — the reference crashes in np.quantile here, data.py:40):
+        # fall back to the un-saturated quantiles, i.e. plain min/max
+        # stretch for that channel.
+        if not (np.isfinite(lo_q) and 0.0 <= lo_q <= 0.5):
+            lo_q, hi_q = 0.0, 1.0
         lo_v, hi_v = np.quantile(flat[ch], [lo_q, hi_q])
         clipped = np.clip(flat[ch], lo_v, hi_v)
         bottom = clipped.min()
