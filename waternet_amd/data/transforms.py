@@ -43,7 +43,8 @@ def white_balance_transform(im_rgb: np.ndarray) -> np.ndarray:
             [im_rgb[:, :, i].astype(np.int64).sum() for i in range(p)],
             dtype=np.float64,
         )
-        ratio = chan_sums.max() / chan_sums
+        with np.errstate(divide="ignore", invalid="ignore"):
+            ratio = chan_sums.max() / chan_sums  # inf/nan handled below
         sat_lo = 0.005 * ratio
         sat_hi = 0.005 * ratio
         flat = im_rgb.reshape(h * w, p).T.astype(np.float64)
