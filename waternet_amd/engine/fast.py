@@ -34,6 +34,7 @@ from waternet_amd.models.waternet import WaterNet
 from waternet_amd.ops.adam import FusedAdam
 from waternet_amd.ops.functional import NormalizeNhwc, mse255_nhwc
 from waternet_amd.ops.preprocess import gpu_transform_batch
+from waternet_amd.utils.streams import DEBUG as STREAM_DEBUG, StreamJoin
 from waternet_amd.ops.ssim import ssim_nhwc
 
 TRAIN_KEYS = ["loss", "perceptual", "mse255", "ssim", "psnr"]
@@ -95,7 +96,10 @@ class FastStepEngine:
         self._steps = 0
         self._eval_batches = 0
         self._graph = None
-        self._use_graph = use_graph
+        # sanitizer mode (WATERNET_AMD_STREAM_DEBUG=1) synchronizes every
+        # cross-stream join, which is illegal inside graph capture — run
+        # eager so misordered reads fail loudly instead of racing
+        self._use_graph = use_graph and not STREAM_DEBUG
         self._comm_stream = torch.cuda.Stream()
         self._vgg_stream = torch.cuda.Stream()
         self._pack_desc = None
@@ -144,18 +148,21 @@ class FastStepEngine:
 
         # The fy tower depends only on ref: run it on a second stream
         # concurrently with the WaterNet forward + fx tower (the deep VGG
-        # layers underfill the 256 CUs, so the towers overlap).
+        # layers underfill the 256 CUs, so the towers overlap). StreamJoin
+        # asserts the producer ordering (sanitizer mode forces completion).
         self._vgg_stream.wait_stream(torch.cuda.current_stream())
+        vgg_join = StreamJoin(self._vgg_stream)
         with torch.cuda.stream(self._vgg_stream), torch.no_grad(), \
                 trace_range("vgg_ref"):
             fy = vgg_forward_nhwc(self.vgg, NormalizeNhwc.apply(ref_nhwc))
+        vgg_join.mark()
 
         with trace_range("forward"):
             out_nhwc = waternet_forward_from_inputs(self.model, *inputs)
 
         with trace_range("loss"):
             fx = vgg_forward_nhwc(self.vgg, NormalizeNhwc.apply(out_nhwc))
-            torch.cuda.current_stream().wait_stream(self._vgg_stream)
+            vgg_join.wait()
             perceptual = mse255_nhwc(fx, fy, 512)
             mse = mse255_nhwc(out_nhwc, ref_nhwc, 3)
             loss = PERCEPTUAL_WEIGHT * perceptual + mse
@@ -172,8 +179,10 @@ class FastStepEngine:
             with trace_range("allreduce"):
                 self.opt.grads.div_(self.world)
                 self._comm_stream.wait_stream(torch.cuda.current_stream())
+                comm_join = StreamJoin(self._comm_stream)
                 with torch.cuda.stream(self._comm_stream):
                     torch.distributed.all_reduce(self.opt.grads)
+                comm_join.mark()
 
         with trace_range("metrics"), torch.no_grad():
             out_d = out_nhwc.detach()
@@ -186,7 +195,7 @@ class FastStepEngine:
             ])
 
         if self.world > 1:
-            torch.cuda.current_stream().wait_stream(self._comm_stream)
+            comm_join.wait()
         with trace_range("optimizer"):
             self.opt.step()
 
