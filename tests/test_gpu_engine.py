@@ -354,3 +354,29 @@ def test_serve_enhance_gpu():
 
     with _pytest.raises(ValueError):
         srv.enhance(big)
+
+
+def test_engine_step_sanitizer_mode(monkeypatch):
+    """One real engine step with WATERNET_AMD_STREAM_DEBUG semantics: every
+    cross-stream join synchronizes (misordered reads become deterministic
+    failures) and graph capture is disabled."""
+    import waternet_amd.engine.fast as fastmod
+    import waternet_amd.utils.streams as streams
+
+    monkeypatch.setattr(streams, "DEBUG", True)
+    monkeypatch.setattr(fastmod, "STREAM_DEBUG", True)
+    from waternet_amd.models.waternet import WaterNet
+
+    torch.manual_seed(4)
+    model = WaterNet().to(DEV)
+    eng = fastmod.FastStepEngine(model, batch_size=2, height=64, width=64,
+                                 device=DEV, use_graph=True)
+    assert eng._use_graph is False  # sanitizer forces eager
+    rng = np.random.default_rng(0)
+    raw = torch.from_numpy(rng.integers(0, 256, (2, 64, 64, 3),
+                                        dtype=np.uint8))
+    eng.load_batch(raw.to(DEV), raw.to(DEV))
+    eng.step()
+    eng.step()
+    torch.cuda.synchronize()
+    assert all(np.isfinite(v) for v in eng.metrics().values())
