@@ -12,16 +12,11 @@ Per frame at steady state: one H2D copy in, one graph replay, one D2H copy
 out. Falls back to eager kernel launches if capture fails.
 """
 
-import os
-
 import numpy as np
 import torch
 
 from waternet_amd.ops import ext
 from waternet_amd.ops.preprocess import gpu_transform_batch
-
-# WN_INFER_PIN=0 disables the pinned-host staging (A/B switch)
-_USE_PIN = os.environ.get("WN_INFER_PIN", "1") == "1"
 
 
 def pad8(rgb_u8: np.ndarray):
@@ -95,21 +90,15 @@ class InferenceEngine:
     def infer_frame(self, rgb_u8: np.ndarray) -> np.ndarray:
         """uint8 HWC RGB frame -> enhanced uint8 HWC RGB."""
         assert rgb_u8.shape == (self.h, self.w, 3)
-        if _USE_PIN:
-            self._in_pin[0].numpy()[:] = rgb_u8  # host memcpy into pinned
-            self.raw_static.copy_(self._in_pin, non_blocking=True)
-        else:
-            self.raw_static.copy_(
-                torch.from_numpy(np.ascontiguousarray(rgb_u8)).unsqueeze(0))
+        self._in_pin[0].numpy()[:] = rgb_u8  # host memcpy into pinned
+        self.raw_static.copy_(self._in_pin, non_blocking=True)
         if self._use_graph and self._graph is None:
             self._capture()
         if self._graph is not None:
             self._graph.replay()
         else:
             self._body()
-        if _USE_PIN:
-            self._out_pin.copy_(self.out_static, non_blocking=True)
-            torch.cuda.current_stream().synchronize()
-            # fresh array: the pinned buffer is reused by the next frame
-            return self._out_pin[0].numpy().copy()
-        return self.out_static[0].cpu().numpy()
+        self._out_pin.copy_(self.out_static, non_blocking=True)
+        torch.cuda.current_stream().synchronize()
+        # fresh array: the pinned buffer is reused by the next frame
+        return self._out_pin[0].numpy().copy()
