@@ -43,12 +43,6 @@ class InferenceEngine:
         self.h, self.w = height, width
         self.raw_static = torch.empty(1, height, width, 3,
                                       dtype=torch.uint8, device=self.device)
-        # pinned host staging: a 1080p frame is ~6 MB each way — pageable
-        # transfers cost ~2-3 ms/frame, pinned ~0.2 ms (plus one host
-        # memcpy into/out of the pinned buffer)
-        self._in_pin = torch.empty(1, height, width, 3, dtype=torch.uint8,
-                                   pin_memory=True)
-        self._out_pin = torch.empty_like(self._in_pin)
         self.out_static = None
         self._graph = None
         self._use_graph = use_graph and height % 8 == 0 and width % 8 == 0
@@ -90,15 +84,13 @@ class InferenceEngine:
     def infer_frame(self, rgb_u8: np.ndarray) -> np.ndarray:
         """uint8 HWC RGB frame -> enhanced uint8 HWC RGB."""
         assert rgb_u8.shape == (self.h, self.w, 3)
-        self._in_pin[0].numpy()[:] = rgb_u8  # host memcpy into pinned
-        self.raw_static.copy_(self._in_pin, non_blocking=True)
+        self.raw_static.copy_(
+            torch.from_numpy(np.ascontiguousarray(rgb_u8)).unsqueeze(0)
+        )
         if self._use_graph and self._graph is None:
             self._capture()
         if self._graph is not None:
             self._graph.replay()
         else:
             self._body()
-        self._out_pin.copy_(self.out_static, non_blocking=True)
-        torch.cuda.current_stream().synchronize()
-        # fresh array: the pinned buffer is reused by the next frame
-        return self._out_pin[0].numpy().copy()
+        return self.out_static[0].cpu().numpy()
