@@ -39,6 +39,7 @@ _ap.add_argument("--bs", type=int, default=8)
 _ap.add_argument("--hw", type=int, default=64)
 _ap.add_argument("--flagship", action="store_true",
                  help="BASELINE config 2 shape: bs=16, 112x112, 400 steps")
+_ap.add_argument("--seed", type=int, default=0)
 _ap.add_argument("--out", default="convergence.json")
 _args = _ap.parse_args()
 if _args.flagship:
@@ -50,7 +51,7 @@ BS, H, W = _args.bs, _args.hw, _args.hw
 
 
 def make_data():
-    rng = np.random.default_rng(0)
+    rng = np.random.default_rng(_args.seed)
     # fixed small synthetic set: smooth ramps + noise so SSIM is meaningful
     yy, xx = np.mgrid[0:H, 0:W]
     base = ((yy + xx) * 255 / (H + W)).astype(np.uint8)
@@ -66,7 +67,7 @@ def make_data():
 
 def run(eager: bool):
     os.environ["WATERNET_AMD_EAGER"] = "1" if eager else "0"
-    torch.manual_seed(0)
+    torch.manual_seed(_args.seed)
     model = WaterNet().to(DEV)
     vgg = PerceptualModel(seed=1234).to(DEV).eval()
     opt = torch.optim.Adam(model.parameters(), lr=1e-3)
