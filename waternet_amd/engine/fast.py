@@ -46,6 +46,30 @@ VAL_KEYS = ["mse", "ssim", "psnr", "perceptual_loss"]
 VGG_SEED = 1234
 
 
+@torch.no_grad()
+def eval_metrics_batch(model, vgg, raw_u8, ref_u8):
+    """GPU preprocess + native forward + the VAL metrics for one uint8
+    device batch. Returns a float64 4-vector in VAL_KEYS order
+    [mse255, ssim, psnr, perceptual] — the building block of both the
+    engine's eval epochs and score.py's fast path."""
+    from waternet_amd.ops import ext
+
+    wb_u8, gc_u8, he_u8 = gpu_transform_batch(raw_u8)
+    e = ext()
+    inputs = e.build_inputs_u8(raw_u8, wb_u8, he_u8, gc_u8)
+    ref_nhwc = e.u8_to_nhwc(ref_u8, 16)
+    fy = vgg_forward_nhwc(vgg, NormalizeNhwc.apply(ref_nhwc))
+    out_nhwc = waternet_forward_from_inputs(model, *inputs)
+    fx = vgg_forward_nhwc(vgg, NormalizeNhwc.apply(out_nhwc))
+    perceptual = mse255_nhwc(fx, fy, 512)
+    mse = mse255_nhwc(out_nhwc, ref_nhwc, 3)
+    ssim = ssim_nhwc(out_nhwc, ref_nhwc, 3, 1.0)
+    mse01 = mse.double() / (255.0 * 255.0)
+    psnr = 10.0 * torch.log10(1.0 / mse01)
+    return torch.stack([mse.double(), ssim.double(), psnr,
+                        perceptual.double()])
+
+
 class FastStepEngine:
     """GPU train/eval step runner over the native CDNA4 kernel library.
 
@@ -277,18 +301,8 @@ class FastStepEngine:
     def eval_batch(self, raw_u8, ref_u8):
         """No-grad forward + metrics on a device uint8 batch; accumulates
         into eval_sums (VAL_KEYS order). Any batch size."""
-        inputs, ref_nhwc = self._build_all_inputs(raw_u8, ref_u8)
-        fy = vgg_forward_nhwc(self.vgg, NormalizeNhwc.apply(ref_nhwc))
-        out_nhwc = waternet_forward_from_inputs(self.model, *inputs)
-        fx = vgg_forward_nhwc(self.vgg, NormalizeNhwc.apply(out_nhwc))
-        perceptual = mse255_nhwc(fx, fy, 512)
-        mse = mse255_nhwc(out_nhwc, ref_nhwc, 3)
-        ssim = ssim_nhwc(out_nhwc, ref_nhwc, 3, 1.0)
-        mse01 = mse.double() / (255.0 * 255.0)
-        psnr = 10.0 * torch.log10(1.0 / mse01)
-        self.eval_sums += torch.stack([
-            mse.double(), ssim.double(), psnr, perceptual.double()
-        ])
+        self.eval_sums += eval_metrics_batch(self.model, self.vgg, raw_u8,
+                                             ref_u8)
         self._eval_batches += 1
 
     def reset_eval(self):
