@@ -1,7 +1,8 @@
 """wgrad v4 debug probes: per-shape parity vs F.conv2d autograd with
 wrong-index structure analysis (which k/c/tap regions are bad), a
-determinism check, and a big-M variant. Run with WN_WGRAD_SPLIT=1 to rule
-out the z-split atomics."""
+determinism check, and a big-M variant. (The WN_WGRAD_SPLIT knob it once
+drove was removed with the reverted ring-wgrad prototype; the probe
+remains useful for any future wgrad work.)"""
 
 import sys
 from pathlib import Path
@@ -73,8 +74,7 @@ def probe(ks, C, K, N, H, W, seed=0):
 def main():
     import os
 
-    print(f"WN_WGRAD_V4={os.environ.get('WN_WGRAD_V4', '(1)')} "
-          f"WN_WGRAD_SPLIT={os.environ.get('WN_WGRAD_SPLIT', '(0)')}")
+    print(f"WN_WGRAD_M32={os.environ.get('WN_WGRAD_M32', '(1)')}")
     for ks, C, K in SHAPES:
         probe(ks, C, K, 2, 16, 16)
     print("--- big M ---")
