@@ -1,4 +1,4 @@
-"""wgrad v4 debug probes: per-shape parity vs F.conv2d autograd with
+"""wgrad debug probes: per-shape parity vs F.conv2d autograd with
 wrong-index structure analysis (which k/c/tap regions are bad), a
 determinism check, and a big-M variant. (The WN_WGRAD_SPLIT knob it once
 drove was removed with the reverted ring-wgrad prototype; the probe
