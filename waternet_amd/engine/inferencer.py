@@ -84,6 +84,8 @@ class InferenceEngine:
     def infer_frame(self, rgb_u8: np.ndarray) -> np.ndarray:
         """uint8 HWC RGB frame -> enhanced uint8 HWC RGB."""
         assert rgb_u8.shape == (self.h, self.w, 3)
+        # from_numpy on a read-only array (e.g. PIL-derived) warns once;
+        # intentional zero-copy — the tensor is only ever READ by copy_
         self.raw_static.copy_(
             torch.from_numpy(np.ascontiguousarray(rgb_u8)).unsqueeze(0)
         )
