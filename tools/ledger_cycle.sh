@@ -11,6 +11,6 @@ python3 - << PYEOF
 import json
 d = json.load(open("/tmp/b.json"))
 d["sclk_mhz_after"] = int("${SCLK:-0}") or None
-open("gpurun_out/ledger_entry.json", "w").write(json.dumps(d))
+open("gpurun_out/ledger_entry.json", "w").write(json.dumps(d) + "\n")
 print(round(d["value"], 1), "img/s, sclk", d["sclk_mhz_after"])
 PYEOF
