@@ -3,8 +3,8 @@
 # context appended to gpurun_out/ledger_entry.json
 set -x
 export HSA_ENABLE_IPC_MODE_LEGACY=0
-SUITE=${1:-tests/test_gpu_engine.py}
-python -m pytest "$SUITE" -q 2>&1 | grep -E "passed|failed"
+SUITES=${@:-tests/test_gpu_engine.py}
+python -m pytest $SUITES -q 2>&1 | grep -E "passed|failed"
 timeout 300 python bench.py --steps 30 --warmup 10 2>/dev/null | tail -1 > /tmp/b.json
 SCLK=$(rocm-smi --showgpuclocks 2>/dev/null | grep -oE '\([0-9]+Mhz\)' | head -1 | tr -d '(Mhz)')
 python3 - << PYEOF
