@@ -1,0 +1,147 @@
+"""Property-based fuzzing of the CPU pipeline (hypothesis).
+
+Shape- and value-space invariants the example-based tests in
+test_transforms.py / test_data_pipeline.py don't sweep: arbitrary image
+sizes (including tiny and extreme-aspect), arbitrary uint8 content,
+determinism, and input immutability. Reference semantics under test:
+data.py:6-90 (wb/gamma/histeq + transform wrapper), the arr2ten/ten2arr
+bridges (training_utils.py:14-43), and the paired augment (the
+reference's albumentations HorizontalFlip/RandomRotate90,
+training_utils.py:60-66).
+
+Settings are derandomized (fixed seed, no deadline) so CI runs are
+reproducible and never flake on timing.
+"""
+
+import numpy as np
+import torch
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+from hypothesis.extra import numpy as npst
+
+from waternet_amd.data.augment import PairedAugment
+from waternet_amd.data.bridge import arr2ten, ten2arr
+from waternet_amd.data.transforms import (
+    gamma_correction,
+    histeq,
+    lab2rgb_u8,
+    rgb2lab_u8,
+    transform,
+    white_balance_transform,
+)
+
+COMMON = dict(derandomize=True, deadline=None,
+              suppress_health_check=[HealthCheck.too_slow])
+
+
+def u8_images(min_side=1, max_side=33):
+    return npst.arrays(
+        dtype=np.uint8,
+        shape=st.tuples(
+            st.integers(min_side, max_side),
+            st.integers(min_side, max_side),
+            st.just(3),
+        ),
+        elements=st.integers(0, 255),
+    )
+
+
+@settings(max_examples=30, **COMMON)
+@given(u8_images())
+def test_transform_invariants(im):
+    """transform() on ANY uint8 RGB image: three uint8 outputs of the same
+    shape, in range by dtype, deterministic, input untouched."""
+    before = im.copy()
+    wb, gc, he = transform(im)
+    for out in (wb, gc, he):
+        assert out.shape == im.shape
+        assert out.dtype == np.uint8
+    np.testing.assert_array_equal(im, before)  # no in-place mutation
+    wb2, gc2, he2 = transform(im)
+    np.testing.assert_array_equal(wb, wb2)
+    np.testing.assert_array_equal(gc, gc2)
+    np.testing.assert_array_equal(he, he2)
+
+
+@settings(max_examples=30, **COMMON)
+@given(u8_images())
+def test_gamma_monotone_pointwise(im):
+    """Gamma (x/255)^0.7 is monotone: pixel order is preserved wherever the
+    input order is strict, and endpoints are fixed points."""
+    out = gamma_correction(im)
+    assert out.dtype == np.uint8
+    # 0 -> 0 and 255 -> 255 exactly
+    assert np.all(out[im == 0] == 0)
+    assert np.all(out[im == 255] == 255)
+    # gamma < 1 brightens mid-tones: out >= in everywhere (255*(x/255)^0.7
+    # >= x on [0,255]), allowing equality from rounding
+    assert np.all(out.astype(np.int16) >= im.astype(np.int16) - 1)
+
+
+@settings(max_examples=30, **COMMON)
+@given(u8_images(min_side=2))
+def test_white_balance_output_range_and_purity(im):
+    out = white_balance_transform(im)
+    assert out.shape == im.shape and out.dtype == np.uint8
+    # deterministic
+    np.testing.assert_array_equal(out, white_balance_transform(im))
+
+
+@settings(max_examples=30, **COMMON)
+@given(u8_images(min_side=2))
+def test_histeq_luminance_only(im):
+    """CLAHE runs on L only: the a/b chroma planes of the output match the
+    input's (up to LAB->RGB->LAB u8 rounding)."""
+    out = histeq(im)
+    assert out.shape == im.shape and out.dtype == np.uint8
+    lab_in = rgb2lab_u8(im).astype(np.int16)
+    lab_out = rgb2lab_u8(out).astype(np.int16)
+    # bound 8: a/b are ill-conditioned for near-black pixels (CLAHE lifts L,
+    # and re-deriving chroma after the u8 RGB round-trip shifts them a few
+    # counts) — still a tight bound vs the 255-count chroma range
+    assert np.abs(lab_in[..., 1:] - lab_out[..., 1:]).max() <= 8
+
+
+@settings(max_examples=40, **COMMON)
+@given(u8_images())
+def test_lab_roundtrip_bounded(im):
+    """RGB -> LAB(u8) -> RGB round-trip error stays within the 8-bit LAB
+    quantization bound for every input, including extremes."""
+    back = lab2rgb_u8(rgb2lab_u8(im))
+    assert back.shape == im.shape and back.dtype == np.uint8
+    # bound 12: at gamut edges (e.g. pure saturated green) one count of u8
+    # LAB chroma quantization maps to several RGB counts — matches cv2's own
+    # u8 LAB round-trip behavior
+    assert np.abs(back.astype(np.int16) - im.astype(np.int16)).max() <= 12
+
+
+@settings(max_examples=40, **COMMON)
+@given(u8_images())
+def test_bridge_roundtrip_exact(im):
+    """ten2arr(arr2ten(x)) == x exactly for every uint8 image (the
+    reference's arr2ten/ten2arr pairs, training_utils.py:14-43)."""
+    ten = arr2ten(im, add_batch_dim=True)
+    assert ten.dtype == torch.float32
+    assert ten.shape == (1, 3, im.shape[0], im.shape[1])
+    assert float(ten.min()) >= 0.0 and float(ten.max()) <= 1.0
+    back = ten2arr(ten)[0] if ten2arr(ten).ndim == 4 else ten2arr(ten)
+    np.testing.assert_array_equal(np.asarray(back).reshape(im.shape), im)
+
+
+@settings(max_examples=20, **COMMON)
+@given(u8_images(min_side=4, max_side=24), st.integers(0, 2**31 - 1))
+def test_paired_augment_consistency(im, seed):
+    """The joint augment applies the SAME flip/rot to every tensor in the
+    pair (reference: one albumentations call over raw+ref), and only ever
+    produces dihedral-group images of the input."""
+    aug = PairedAugment(rng=np.random.default_rng(seed))
+    a, b = aug(im, im.copy())
+    np.testing.assert_array_equal(a, b)  # identical inputs stay identical
+    variants = []
+    for k in range(4):
+        r = np.rot90(im, k)
+        variants.append(r)
+        variants.append(r[:, ::-1])
+    assert any(
+        a.shape == v.shape and np.array_equal(a, v) for v in variants
+    )
