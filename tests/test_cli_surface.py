@@ -1,0 +1,105 @@
+"""CPU tests for the remaining user-facing CLI/API surface: directory
+inference with --show-split (reference inference.py:128-135, 202-233), the
+hubconf pretrained offline error path (reference hubconf.py:78-83), and the
+video-IO-without-ffmpeg error (reference uses cv2.VideoCapture,
+inference.py:238-256)."""
+
+import subprocess
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def test_inference_cli_dir_show_split(tmp_path):
+    """A directory source enhances every image; --show-split writes the
+    before/after composite (white divider at w//2)."""
+    from PIL import Image
+
+    rng = np.random.default_rng(0)
+    srcdir = tmp_path / "imgs"
+    srcdir.mkdir()
+    for name in ("a.png", "b.jpg"):
+        img = rng.integers(0, 256, size=(40, 56, 3), dtype=np.uint8)
+        Image.fromarray(img).save(srcdir / name)
+    (srcdir / "notes.txt").write_text("ignored: not an image suffix")
+
+    out = subprocess.run(
+        [sys.executable, "inference.py", "--source", str(srcdir),
+         "--name", "pytest-split", "--show-split"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr
+    outdir = REPO / "output" / "pytest-split"
+    for name in ("a.png", "b.jpg"):
+        p = outdir / name
+        assert p.exists(), f"missing {p}"
+        with Image.open(p) as im:
+            arr = np.asarray(im.convert("RGB"))
+        assert arr.shape == (40, 56, 3)
+        # the composite's vertical divider at w//2 is near-white (JPEG
+        # compression smears it slightly on the .jpg output)
+        assert arr[:, 56 // 2].mean() >= 200, "split divider missing"
+    assert "Total images/videos: 2" in out.stdout
+
+
+def test_hubconf_pretrained_offline_error(monkeypatch):
+    """pretrained=True without network raises a clear, actionable error
+    (mentioning checkpoint=/pretrained=False), not a bare URLError."""
+    import torch
+
+    sys.path.insert(0, str(REPO))
+    try:
+        import hubconf
+    finally:
+        sys.path.pop(0)
+
+    def fail_download(*a, **k):
+        raise OSError("simulated: no route to host")
+
+    monkeypatch.setattr(torch.hub, "load_state_dict_from_url", fail_download)
+    with pytest.raises(RuntimeError) as exc:
+        hubconf.waternet(pretrained=True, device="cpu")
+    msg = str(exc.value)
+    assert "checkpoint=" in msg and "pretrained=False" in msg
+
+
+def test_hubconf_pretrained_local_checkpoint(tmp_path):
+    """checkpoint=<path> loads without network and returns the working
+    (preprocess, postprocess, model) tuple."""
+    import torch
+
+    from waternet_amd.models.waternet import WaterNet
+
+    sys.path.insert(0, str(REPO))
+    try:
+        import hubconf
+    finally:
+        sys.path.pop(0)
+
+    ckpt = tmp_path / "w.pt"
+    torch.save(WaterNet().state_dict(), ckpt)
+    preprocess, postprocess, model = hubconf.waternet(
+        pretrained=True, device="cpu", checkpoint=str(ckpt))
+    rgb = np.random.default_rng(0).integers(
+        0, 256, size=(32, 32, 3), dtype=np.uint8)
+    tens = preprocess(rgb)
+    out = model(*tens)
+    arr = postprocess(out)
+    # batched NHWC, as the reference's ten2arr (training_utils.py:31-43)
+    assert arr.shape == (1, 32, 32, 3) and arr.dtype == np.uint8
+
+
+def test_video_requires_ffmpeg_clear_error(monkeypatch, tmp_path):
+    """Without the ffmpeg binary, video IO raises an actionable error
+    instead of a FileNotFoundError deep in subprocess."""
+    import shutil
+
+    from waternet_amd.engine import video
+
+    monkeypatch.setattr(shutil, "which", lambda name: None)
+    with pytest.raises(RuntimeError, match="ffmpeg"):
+        video.FFmpegReader(tmp_path / "clip.mp4")
