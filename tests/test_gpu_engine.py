@@ -404,36 +404,3 @@ def test_score_cli_fast_vs_eager(tmp_path):
     # it must agree too
     a, b = m_fast["perceptual_loss"], m_eager["perceptual_loss"]
     assert abs(a - b) / (abs(b) + 1e-6) < 0.10, (a, b)
-
-
-def test_banded_inference_bit_exact():
-    """Band-wise forward (L3-residency path for large frames) is BIT-exact
-    vs the whole-frame native forward: the per-pixel FMA order of the conv
-    kernels is band-independent, and halo=13 covers the receptive field
-    (CPU-side proof in test_banded_inference.py). Covers both eager and
-    hipGraph execution and a ragged last band (160 % 48 != 0)."""
-    from waternet_amd.engine.inferencer import InferenceEngine
-    from waternet_amd.models.waternet import WaterNet
-
-    torch.manual_seed(7)
-    model = WaterNet().to(DEV)
-    rng = np.random.default_rng(7)
-    frame = rng.integers(0, 256, size=(160, 192, 3), dtype=np.uint8)
-
-    whole = InferenceEngine(model, 160, 192, device=DEV, use_graph=False,
-                            band_rows=0)
-    banded = InferenceEngine(model, 160, 192, device=DEV, use_graph=False,
-                             band_rows=48)
-    out_w = whole.infer_frame(frame)
-    out_b = banded.infer_frame(frame)
-    assert np.array_equal(out_w, out_b), (
-        f"banded/whole mismatch: max diff "
-        f"{np.abs(out_w.astype(int) - out_b.astype(int)).max()}")
-
-    # same through hipGraph capture
-    banded_g = InferenceEngine(model, 160, 192, device=DEV, use_graph=True,
-                               band_rows=48)
-    out_bg = banded_g.infer_frame(frame)
-    out_bg2 = banded_g.infer_frame(frame)  # replay steady state
-    assert np.array_equal(out_w, out_bg)
-    assert np.array_equal(out_w, out_bg2)

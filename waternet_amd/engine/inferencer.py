@@ -37,26 +37,10 @@ def pad8(rgb_u8: np.ndarray):
 
 class InferenceEngine:
     def __init__(self, model, height, width, device="cuda:0",
-                 use_graph=True, band_rows=None):
-        """band_rows: horizontal banding of the conv towers for
-        L3-residency at large resolutions (see
-        native.waternet_forward_banded). None = auto by working-set size
-        (whole-frame below ~1 Mpixel, ~136-row bands at 1080p); 0 = always
-        whole-frame; >0 = forced band height. WATERNET_AMD_BAND_ROWS
-        overrides (A/B switch)."""
-        import os
-
-        from waternet_amd.engine.native import auto_band_rows
-
+                 use_graph=True):
         self.model = model.eval()
         self.device = torch.device(device)
         self.h, self.w = height, width
-        env = os.environ.get("WATERNET_AMD_BAND_ROWS")
-        if env is not None:
-            band_rows = int(env)
-        if band_rows is None:
-            band_rows = auto_band_rows(height, width)
-        self.band_rows = band_rows
         self.raw_static = torch.empty(1, height, width, 3,
                                       dtype=torch.uint8, device=self.device)
         self.out_static = None
@@ -65,10 +49,7 @@ class InferenceEngine:
 
     @torch.no_grad()
     def _body(self):
-        from waternet_amd.engine.native import (
-            waternet_forward_banded,
-            waternet_forward_from_inputs,
-        )
+        from waternet_amd.engine.native import waternet_forward_from_inputs
 
         raw = self.raw_static
         wb, gc, he = gpu_transform_batch(raw)
@@ -77,11 +58,7 @@ class InferenceEngine:
         # full-NHWC: fused uint8 cat-fold -> convs -> fused u8 postprocess,
         # no NCHW tensor anywhere in the frame pipeline.
         inputs = e.build_inputs_u8(raw, wb, he, gc)
-        if self.band_rows > 0:
-            out_nhwc = waternet_forward_banded(self.model, *inputs,
-                                               band_rows=self.band_rows)
-        else:
-            out_nhwc = waternet_forward_from_inputs(self.model, *inputs)
+        out_nhwc = waternet_forward_from_inputs(self.model, *inputs)
         self.out_static = e.out_to_u8(out_nhwc)
 
     def _capture(self):

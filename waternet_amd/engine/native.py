@@ -76,58 +76,6 @@ def waternet_forward_from_inputs(model, cmg_in, rwb_in, rce_in, rgc_in):
     return GatedFusion.apply(maps, refined[0], refined[1], refined[2])
 
 
-# WaterNet receptive-field radius in rows/cols: the deepest path is the CMG
-# tower (k7,k5,k3,k1,k7,k5,k3,k3 -> 3+2+1+0+3+2+1+1); the refiners reach 6
-# and the fusion is pointwise. With this halo, band-wise execution is
-# BIT-exact vs whole-frame (same per-pixel FMA sequence; zero-padding only
-# ever lands where the crop discards it or where the true image edge is).
-RECEPTIVE_HALO = 13
-
-
-def waternet_forward_banded(model, cmg_in, rwb_in, rce_in, rgc_in,
-                            band_rows):
-    """Band-wise NHWC forward for LARGE single-image inference: run the conv
-    towers on horizontal bands of `band_rows` rows (+RECEPTIVE_HALO each
-    side) so each band's activations stay resident in the 256 MB Infinity
-    Cache instead of spilling the k5/k7 halo tap re-reads to HBM3E.
-
-    At 1080p the whole-frame CMG activations are ~1 GB live (two 128-ch
-    bf16 tensors x 2.1 M pixels), so every tap re-read is HBM traffic
-    (measured FETCH 9-18x unique bytes at 512^2 — profiles/
-    r02_512sq_roofline.json); a 128-row band's working set is ~150 MB.
-
-    N must be 1 (row slices of an NHWC batch are only contiguous for a
-    single image). Returns the same (1,H,W,16) bf16 tensor as
-    waternet_forward_from_inputs, bit-identical.
-    """
-    n, h, _, _ = cmg_in.shape
-    assert n == 1, "banded forward is a single-image (inference) path"
-    inputs = (cmg_in, rwb_in, rce_in, rgc_in)
-    outs = []
-    r0 = 0
-    while r0 < h:
-        r1 = min(r0 + band_rows, h)
-        hs = max(r0 - RECEPTIVE_HALO, 0)
-        he = min(r1 + RECEPTIVE_HALO, h)
-        band = [t[:, hs:he] for t in inputs]  # contiguous views (N=1)
-        ob = waternet_forward_from_inputs(model, *band)
-        outs.append(ob[:, r0 - hs: r0 - hs + (r1 - r0)])
-        r0 = r1
-    return torch.cat(outs, dim=1)
-
-
-def auto_band_rows(h, w, target_bytes=160 << 20):
-    """0 = run whole-frame (live set already cache-resident); otherwise the
-    band height whose working set (two 128-ch bf16 NHWC tensors of
-    band+2*halo rows) fits in `target_bytes` of the 256 MB L3 (headroom
-    left for the inputs, weights and refiner tensors)."""
-    per_row = w * (128 + 128) * 2
-    if h * per_row <= target_bytes:
-        return 0
-    rows = int(target_bytes // per_row) - 2 * RECEPTIVE_HALO
-    return max(rows, 32)
-
-
 def waternet_forward_native(model, x, wb, ce, gc):
     """x, wb, ce, gc: (N,3,H,W) fp32 CUDA -> (N,3,H,W) fp32.
 
