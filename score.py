@@ -66,11 +66,18 @@ def main(argv=None):
             raw_mode=use_fast,
         )
         split = [800, 90]
+    if len(dataset) != sum(split):
+        raise SystemExit(
+            f"the UIEB 800/90 split (reference train.py:233) needs exactly "
+            f"{sum(split)} images; found {len(dataset)} under "
+            f"{args.data_root}")
     _, val_dataset = torch.utils.data.random_split(dataset, split)
     val_loader = torch.utils.data.DataLoader(val_dataset,
                                              batch_size=args.batch_size)
 
     model = WaterNet()
+    if not Path(args.weights).is_file():
+        raise SystemExit(f"weights file not found: {args.weights}")
     with open(args.weights, "rb") as f:
         model.load_state_dict(torch.load(f, map_location="cpu"))
     model.to(device).eval()

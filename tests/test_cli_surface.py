@@ -103,3 +103,54 @@ def test_video_requires_ffmpeg_clear_error(monkeypatch, tmp_path):
     monkeypatch.setattr(shutil, "which", lambda name: None)
     with pytest.raises(RuntimeError, match="ffmpeg"):
         video.FFmpegReader(tmp_path / "clip.mp4")
+
+
+def test_score_missing_weights_clear_error():
+    out = subprocess.run(
+        [sys.executable, "score.py", "--weights", "/nonexistent/w.pt",
+         "--synthetic", "4", "--batch-size", "2",
+         "--height", "32", "--width", "32"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "weights file not found" in out.stderr
+
+
+def test_train_missing_dataset_clear_error(tmp_path):
+    out = subprocess.run(
+        [sys.executable, str(REPO / "train.py"), "--data-root",
+         str(tmp_path / "absent"), "--epochs", "1"],
+        cwd=tmp_path, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "raw-890" in out.stderr  # names the expected layout
+
+
+def test_train_resume_missing_savedir_clear_error(tmp_path):
+    out = subprocess.run(
+        [sys.executable, str(REPO / "train.py"), "--synthetic", "4",
+         "--batch-size", "2", "--height", "32", "--width", "32",
+         "--epochs", "1", "--resume", str(tmp_path / "absent")],
+        cwd=tmp_path, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "no last.pt" in out.stderr
+
+
+def test_uieb_dataset_wrong_size_split_error(tmp_path):
+    """A dataset dir with the wrong image count gets the explicit 800/90
+    message, not torch's opaque random_split ValueError."""
+    from PIL import Image
+
+    (tmp_path / "raw-890").mkdir()
+    (tmp_path / "reference-890").mkdir()
+    img = np.zeros((40, 40, 3), dtype=np.uint8)
+    for d in ("raw-890", "reference-890"):
+        Image.fromarray(img).save(tmp_path / d / "0.png")
+    out = subprocess.run(
+        [sys.executable, str(REPO / "train.py"), "--data-root",
+         str(tmp_path), "--epochs", "1"],
+        cwd=tmp_path, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "800/90" in out.stderr and "found 1" in out.stderr

@@ -211,6 +211,11 @@ def main(argv=None):
             raw_mode=use_fast,
         )
         split = [800, 90]
+    if len(dataset) != sum(split):
+        raise SystemExit(
+            f"the UIEB 800/90 split (reference train.py:233) needs exactly "
+            f"{sum(split)} images; found {len(dataset)} under "
+            f"{args.data_root} (use --synthetic N for on-the-fly data)")
     train_dataset, val_dataset = torch.utils.data.random_split(dataset, split)
     train_dataset = shard_dataset(train_dataset, dist_env)
 
@@ -236,8 +241,12 @@ def main(argv=None):
 
     model = WaterNet()
     if args.resume is not None:
-        model.load_state_dict(torch.load(Path(args.resume) / "last.pt",
-                                         map_location="cpu"))
+        ckpt = Path(args.resume) / "last.pt"
+        if not ckpt.is_file():
+            raise SystemExit(
+                f"--resume {args.resume}: no last.pt there — pass the "
+                "savedir of a previous run (e.g. training/0)")
+        model.load_state_dict(torch.load(ckpt, map_location="cpu"))
     elif args.weights is not None:
         with open(args.weights, "rb") as f:
             model.load_state_dict(torch.load(f, map_location="cpu"))

@@ -54,12 +54,23 @@ class UIEBDataset(torch.utils.data.Dataset):
         """raw_mode=True: skip the CPU preprocess transforms and return
         {raw, ref} uint8 HWC tensors — the fast GPU-preprocess train path
         runs wb/gamma/clahe on-device inside the step instead."""
+        for d in (raw_dir, ref_dir):
+            if not Path(d).is_dir():
+                raise FileNotFoundError(
+                    f"UIEB dataset directory not found: {d} — expected a "
+                    "root containing raw-890/ and reference-890/ of paired "
+                    ".png images (pass --data-root, or --synthetic N to "
+                    "train without a dataset on disk)")
         raw_im_fns = sorted(
             p.name for p in Path(raw_dir).glob("*.png")
         )
         ref_im_fns = sorted(
             p.name for p in Path(ref_dir).glob("*.png")
         )
+        if not raw_im_fns:
+            raise FileNotFoundError(
+                f"no .png images in {raw_dir} — expected the UIEB raw-890 "
+                "set (or use --synthetic N)")
         assert set(raw_im_fns) == set(ref_im_fns), "raw/ref file mismatch"
 
         self.transform = transform if transform is not None else PairedAugment()
