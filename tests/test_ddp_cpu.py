@@ -223,13 +223,18 @@ def _run_arena_ddp_worker(rank, world_size, port, q):
         q.put((rank, "ERROR", repr(e)))
 
 
-def test_bench_arena_ddp_flow_world2():
-    world_size = 2
+@pytest.mark.parametrize("world_size,port", [(2, 29817), (4, 29818),
+                                             (8, 29819)])
+def test_bench_arena_ddp_flow(world_size, port):
+    """The driver's scale bench runs this exact arithmetic at
+    --nproc-per-node 8; world 8 on CPU/gloo proves the 8-rank flow
+    (per-rank init -> rank-0 master broadcast -> grads/8 + SUM all-reduce
+    -> step keeps all ranks bit-identical) before any 8-GPU hardware."""
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [
         ctx.Process(target=_run_arena_ddp_worker,
-                    args=(r, world_size, 29817, q))
+                    args=(r, world_size, port, q))
         for r in range(world_size)
     ]
     for p in procs:
@@ -243,4 +248,5 @@ def test_bench_arena_ddp_flow_world2():
     sums = {round(r[1], 9) for r in results}
     assert len(sums) == 1, results
     w = {r[0]: torch.tensor(r[2]) for r in results}
-    assert torch.equal(w[0], w[1])
+    for r in range(1, world_size):
+        assert torch.equal(w[0], w[r]), f"rank {r} diverged"
