@@ -187,3 +187,35 @@ def test_eval_reference_bug_mode(monkeypatch):
     # other metrics must be identical
     assert abs(fixed["perceptual_loss"] - buggy["perceptual_loss"]) > 1e-9
     assert abs(fixed["mse"] - buggy["mse"]) < 1e-9
+
+
+def test_metrics_csv_byte_format(tmp_path, monkeypatch):
+    """The CSV/config.json output format is the reference's exactly
+    (train.py:311-348): np.savetxt fmt=%f comma-delimited with the bare
+    header line, and config.json with the same 5 keys."""
+    import json
+    import re
+
+    import train as train_cli
+
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setattr(train_cli, "__file__", str(tmp_path / "train.py"))
+    train_cli.main(["--epochs", "2", "--batch-size", "2", "--height", "32",
+                    "--width", "32", "--synthetic", "6"])
+    savedir = tmp_path / "training" / "0"
+
+    tr = (savedir / "metrics-train.csv").read_text().splitlines()
+    assert tr[0] == "mse,ssim,psnr,perceptual_loss,loss"
+    assert len(tr) == 3  # header + one row per epoch
+    row = re.compile(r"^-?\d+\.\d{6}(,-?\d+\.\d{6}){4}$")  # %f x5
+    assert all(row.match(l) for l in tr[1:]), tr[1:]
+
+    va = (savedir / "metrics-val.csv").read_text().splitlines()
+    assert va[0] == "mse,ssim,psnr,perceptual_loss"
+    vrow = re.compile(r"^-?\d+\.\d{6}(,-?\d+\.\d{6}){3}$")
+    assert all(vrow.match(l) for l in va[1:]), va[1:]
+
+    cfg = json.loads((savedir / "config.json").read_text())
+    assert set(cfg) == {"epochs", "batch_size", "im_height", "im_width",
+                        "weights"}
+    assert cfg["epochs"] == 2 and cfg["im_height"] == 32
