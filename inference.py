@@ -41,16 +41,16 @@ def parse_args(argv=None):
 
 
 def make_savedir(name):
+    """Pick output/<name or next number>. The directory itself is created
+    as late as possible — at the first write — so failed runs leave no
+    empty savedirs (reference inference.py:198-200)."""
     outputdir = Path("output")
-    outputdir.mkdir(exist_ok=True)
     if name is not None:
-        savedir = outputdir / name
-    else:
-        nums = [int(p.stem) for p in outputdir.glob("*")
-                if p.is_dir() and p.stem.isdecimal()]
-        savedir = outputdir / str(max(nums) + 1 if nums else 0)
-    savedir.mkdir(exist_ok=True)
-    return savedir
+        return outputdir / name
+    nums = [int(p.stem) for p in outputdir.glob("*")
+            if p.is_dir() and p.stem.isdecimal()] if outputdir.is_dir() \
+        else []
+    return outputdir / str(max(nums) + 1 if nums else 0)
 
 
 def load_model(weights, device):
@@ -150,6 +150,7 @@ def run_image(model, path: Path, savedir: Path, device, show_split):
         rgb = np.asarray(im.convert("RGB"))
     out = enhance_frame(model, rgb, device)
     result = compose_split(rgb, out) if show_split else out
+    savedir.mkdir(parents=True, exist_ok=True)  # late, ref inference.py:198
     Image.fromarray(result).save(savedir / path.name)
 
 
@@ -157,6 +158,7 @@ def run_video(model, path: Path, savedir: Path, device, show_split):
     from waternet_amd.engine.video import FFmpegReader, FFmpegWriter
 
     reader = FFmpegReader(path)
+    savedir.mkdir(parents=True, exist_ok=True)  # late, ref inference.py:198
     outpath = savedir / path.name
     writer = FFmpegWriter(outpath, reader.width, reader.height, reader.fps)
     n = 0

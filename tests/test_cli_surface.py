@@ -154,3 +154,33 @@ def test_uieb_dataset_wrong_size_split_error(tmp_path):
     )
     assert out.returncode != 0
     assert "800/90" in out.stderr and "found 1" in out.stderr
+
+
+def test_inference_failed_run_leaves_no_savedir(tmp_path):
+    """Reference behavior (inference.py:198-200): the savedir is created at
+    the first write, so a run that fails early leaves no empty output
+    dir; successful numbered runs count up from 0."""
+    from PIL import Image
+
+    img = np.zeros((16, 16, 3), dtype=np.uint8)
+    src = tmp_path / "x.png"
+    Image.fromarray(img).save(src)
+
+    # failing run: bad --weights aborts before any write
+    out = subprocess.run(
+        [sys.executable, str(REPO / "inference.py"), "--source", str(src),
+         "--weights", str(tmp_path / "missing.pt")],
+        cwd=tmp_path, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert not (tmp_path / "output").exists()
+
+    # successful runs: output/0 then output/1
+    for expect in ("0", "1"):
+        out = subprocess.run(
+            [sys.executable, str(REPO / "inference.py"),
+             "--source", str(src)],
+            cwd=tmp_path, capture_output=True, text=True, timeout=600,
+        )
+        assert out.returncode == 0, out.stderr
+        assert (tmp_path / "output" / expect / "x.png").exists()
