@@ -96,7 +96,13 @@ def create_app(weights: Optional[str] = None, device: Optional[str] = None):
         body = await request.body()
         if len(body) > MAX_BODY_BYTES:
             return Response(status_code=413, content="image body too large")
-        img = Image.open(io.BytesIO(body)).convert("RGB")
+        try:
+            # PIL's MAX_IMAGE_PIXELS decompression-bomb guard applies here;
+            # an oversized or undecodable body is a client error, not a 500
+            img = Image.open(io.BytesIO(body)).convert("RGB")
+        except Exception as e:  # noqa: BLE001
+            return Response(status_code=400,
+                            content=f"body is not a decodable image: {e}")
         rgb = np.asarray(img)
         try:
             out = server.enhance(rgb)

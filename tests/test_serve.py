@@ -53,3 +53,9 @@ def test_serve_gpu_engine():
     assert out1.shape == img.shape
     assert (out1 == out2).all()
     assert (64, 80) in s._engines
+
+
+def test_enhance_rejects_garbage_body(client):
+    r = client.post("/enhance", content=b"this is not an image")
+    assert r.status_code == 400
+    assert "not a decodable image" in r.text
