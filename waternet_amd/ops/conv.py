@@ -1,4 +1,6 @@
-"""Conv2d (stride-1, same-pad) over the CDNA4 MFMA implicit-GEMM kernels.
+"""Conv2d (stride-1, same-pad) over the CDNA4 MFMA implicit-GEMM kernels —
+every nn.Conv2d the reference runs (net.py:14-45,66-71 and the VGG19
+features) dispatches here on GPU.
 
 ConvBiasAct is the autograd wrapper around csrc/conv_mfma.hip: forward runs
 the fused conv+bias+activation kernel on NHWC bf16; backward runs

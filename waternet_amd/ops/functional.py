@@ -1,4 +1,10 @@
-"""Autograd wrappers for the fused elementwise / layout / reduction kernels."""
+"""Autograd wrappers for the fused elementwise / layout / reduction kernels.
+
+Reference parity: GatedFusion = net.py:104-108 (K15),
+NormalizeNhwc/NormalizeVgg = the TF.normalize calls in train.py:119-123
+(K16), mse255_nhwc/Mse255 = the 255-scale MSE reductions in
+train.py:125-131 (K18/K19), MaxPool2x2 = the VGG feature maxpools (K17),
+NhwcToNchw/NchwToNhwc = layout bridges for the public NCHW contract."""
 
 import torch
 

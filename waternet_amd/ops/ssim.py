@@ -1,4 +1,6 @@
-"""Native HIP SSIM (metric path, no autograd)."""
+"""Native HIP SSIM (metric path, no autograd) — the reference's
+torchmetrics StructuralSimilarityIndexMeasure call site (train.py:44-47,
+SURVEY.md §2.2 K21), same 11x11/sigma-1.5 gaussian-window semantics."""
 
 import torch
 
