@@ -184,3 +184,27 @@ def test_inference_failed_run_leaves_no_savedir(tmp_path):
         )
         assert out.returncode == 0, out.stderr
         assert (tmp_path / "output" / expect / "x.png").exists()
+
+
+def test_example_notebook_executes():
+    """examples/example-notebook.ipynb (the reference Colab-example analog)
+    stays valid JSON and its code cells execute top to bottom on CPU."""
+    import json
+
+    nb = json.loads(
+        (REPO / "examples" / "example-notebook.ipynb").read_text())
+    assert nb["nbformat"] == 4
+    ns = {}
+    cwd = REPO / "examples"
+    import os
+
+    old = os.getcwd()
+    os.chdir(cwd)
+    try:
+        for cell in nb["cells"]:
+            if cell["cell_type"] != "code":
+                continue
+            exec(compile("".join(cell["source"]), "<cell>", "exec"), ns)
+    finally:
+        os.chdir(old)
+    assert ns["out_im"].shape == (1, 480, 720, 3)
