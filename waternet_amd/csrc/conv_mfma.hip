@@ -1372,7 +1372,13 @@ void launch_conv_bn(const at::Tensor& x, const at::Tensor& wp,
     } else {
       bool done8 = false;
       if constexpr (BN >= 64) {
-        if (M >= 16384) {  // big-M: 8-wave 256-row phase-split kernel
+        static const long m8_thresh = [] {
+          // A/B override: the M threshold above which the 8-wave 256-row
+          // igemm8 is used instead of the 4-wave 128-row igemm
+          const char* e = getenv("WN_IGEMM8_MTHRESH");
+          return e ? atol(e) : 16384L;
+        }();
+        if (M >= m8_thresh) {  // big-M: 8-wave 256-row phase-split kernel
           static const int var = [] {
             const char* e = getenv("WN_IGEMM8_VAR");
             return e ? atoi(e) : 0;
