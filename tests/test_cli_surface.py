@@ -208,3 +208,48 @@ def test_example_notebook_executes():
     finally:
         os.chdir(old)
     assert ns["out_im"].shape == (1, 480, 720, 3)
+
+
+def test_video_pipeline_with_fake_ffmpeg(tmp_path):
+    """End-to-end video inference (reference inference.py:238-323) with a
+    stub ffmpeg/ffprobe on PATH: the reader parses probe JSON and streams
+    raw RGB frames, every frame runs through the model, and the writer
+    receives raw RGB back — closing the only otherwise-untestable-offline
+    subsystem."""
+    import os
+    import stat
+
+    h, w, frames = 6, 8, 2
+    fake = tmp_path / "bin"
+    fake.mkdir()
+    (fake / "ffprobe").write_text(
+        "#!/bin/bash\n"
+        f'echo \'{{"streams":[{{"width":{w},"height":{h},'
+        '"r_frame_rate":"24/1"}]}\'\n')
+    (fake / "ffmpeg").write_text(
+        "#!/bin/bash\n"
+        'prev=""; input=""\n'
+        'for a in "$@"; do if [ "$prev" = "-i" ]; then input="$a"; fi; '
+        'prev="$a"; done\n'
+        'last="${@: -1}"\n'
+        '# decode mode: -i <file> ... -   |   encode mode: -i - ... <file>\n'
+        'if [ "$input" = "-" ]; then cat - > "$last"; else cat "$input"; fi\n')
+    for f in ("ffprobe", "ffmpeg"):
+        os.chmod(fake / f, stat.S_IRWXU)
+
+    rng = np.random.default_rng(3)
+    raw = rng.integers(0, 256, size=(frames, h, w, 3), dtype=np.uint8)
+    clip = tmp_path / "clip.mp4"
+    clip.write_bytes(raw.tobytes())
+
+    env = dict(os.environ, PATH=f"{fake}:{os.environ['PATH']}")
+    out = subprocess.run(
+        [sys.executable, str(REPO / "inference.py"), "--source", str(clip)],
+        cwd=tmp_path, capture_output=True, text=True, timeout=600, env=env,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "Wrote 2 frames" in out.stdout
+    produced = (tmp_path / "output" / "0" / "clip.mp4").read_bytes()
+    got = np.frombuffer(produced, dtype=np.uint8).reshape(frames, h, w, 3)
+    # frames passed through the actual model: right shape, not a copy
+    assert not np.array_equal(got, raw)
