@@ -219,3 +219,21 @@ def test_metrics_csv_byte_format(tmp_path, monkeypatch):
     assert set(cfg) == {"epochs", "batch_size", "im_height", "im_width",
                         "weights"}
     assert cfg["epochs"] == 2 and cfg["im_height"] == 32
+
+
+def test_resume_of_finished_run_is_noop(tmp_path, monkeypatch):
+    """Resuming a run whose epochs are already complete exits cleanly
+    instead of crashing on empty metric arrays."""
+    import train as train_cli
+
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.setattr(train_cli, "__file__", str(tmp_path / "train.py"))
+    train_cli.main(["--epochs", "1", "--batch-size", "2", "--height", "32",
+                    "--width", "32", "--synthetic", "4", "--full-state"])
+    savedir = tmp_path / "training" / "0"
+    # resume with the same --epochs: start_epoch == epochs -> clean no-op
+    train_cli.main(["--epochs", "1", "--batch-size", "2", "--height", "32",
+                    "--width", "32", "--synthetic", "4", "--full-state",
+                    "--resume", str(savedir)])
+    # the no-op run saved nothing new
+    assert not (tmp_path / "training" / "1" / "metrics-train.csv").exists()

@@ -365,6 +365,14 @@ def main(argv=None):
                     savedir / "last-trainstate.pt",
                 )
 
+    if not saved_train[TRAIN_METRICS_NAMES[0]]:
+        # zero epochs ran (e.g. --resume of an already-finished run, or
+        # --epochs 0): nothing to save — avoid np.stack([]) crashing
+        if rank0:
+            print(f"No epochs to run (start epoch {start_epoch} >= "
+                  f"{args.epochs}); nothing saved")
+        return
+
     if rank0:
         savedir.mkdir(exist_ok=True)
         train_arr = np.stack(
