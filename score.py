@@ -29,6 +29,10 @@ def parse_args(argv=None):
                         default="auto",
                         help="fast = GPU preprocess + native eval (same "
                              "metrics; minutes -> seconds on real datasets)")
+    parser.add_argument("--num-workers", type=int, default=0,
+                        help="DataLoader workers for the val loader "
+                             "(parallelizes the CPU preprocess on the "
+                             "eager engine; reference default 0)")
     return parser.parse_args(argv)
 
 
@@ -73,7 +77,8 @@ def main(argv=None):
             f"{args.data_root}")
     _, val_dataset = torch.utils.data.random_split(dataset, split)
     val_loader = torch.utils.data.DataLoader(val_dataset,
-                                             batch_size=args.batch_size)
+                                             batch_size=args.batch_size,
+                                             num_workers=args.num_workers)
 
     model = WaterNet()
     if not Path(args.weights).is_file():
