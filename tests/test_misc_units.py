@@ -1,0 +1,70 @@
+"""Small-unit CPU tests: the pad8 inference helper, the --show-split
+composite, the bare (unbatched) tensor bridges, and the torchrun env
+plumbing of the parallel layer."""
+
+import numpy as np
+import torch
+
+from waternet_amd.data.bridge import arr2ten, ten2arr
+from waternet_amd.engine.inferencer import pad8
+from waternet_amd.parallel import DistEnv, distributed_env
+
+
+def test_pad8_identity_when_divisible():
+    im = np.zeros((16, 24, 3), dtype=np.uint8)
+    out, h, w = pad8(im)
+    assert out is im and (h, w) == (16, 24)
+
+
+def test_pad8_reflect_and_crop_roundtrip():
+    rng = np.random.default_rng(0)
+    im = rng.integers(0, 256, size=(13, 21, 3), dtype=np.uint8)
+    out, h, w = pad8(im)
+    assert (h, w) == (13, 21)
+    assert out.shape == (16, 24, 3)  # next multiples of 8
+    np.testing.assert_array_equal(out[:13, :21], im)  # crop restores
+    # reflect semantics on the first padded row: mirror of row h-2
+    np.testing.assert_array_equal(out[13, :21], im[11])
+
+
+def test_compose_split_halves():
+    """Left half = original, right half = enhanced (divider/text aside) —
+    reference inference.py:202-233."""
+    from inference import compose_split
+
+    before = np.zeros((40, 60, 3), dtype=np.uint8)
+    after = np.full((40, 60, 3), 200, dtype=np.uint8)
+    comp = compose_split(before, after)
+    assert comp.shape == (40, 60, 3)
+    # below the text region, away from the divider line
+    assert (comp[30:, :28] == 0).all()
+    assert (comp[30:, 33:] == 200).all()
+
+
+def test_bridge_unbatched_paths():
+    rng = np.random.default_rng(1)
+    im = rng.integers(0, 256, size=(9, 7, 3), dtype=np.uint8)
+    ten = arr2ten(im)  # no batch dim: (3, 9, 7)
+    assert ten.shape == (3, 9, 7)
+    back = ten2arr(ten)
+    assert back.shape == (9, 7, 3)
+    np.testing.assert_array_equal(back, im)
+    # ten2arr truncates (not rounds): 0.9999 * 255 = 254.97 -> 254
+    t = torch.full((3, 2, 2), 0.9999)
+    assert ten2arr(t).max() == 254
+
+
+def test_distributed_env_parsing(monkeypatch):
+    monkeypatch.setenv("RANK", "3")
+    monkeypatch.setenv("LOCAL_RANK", "1")
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    env = distributed_env()
+    assert (env.rank, env.local_rank, env.world_size) == (3, 1, 8)
+    assert not env.initialized
+
+
+def test_average_metrics_world1_passthrough():
+    env = DistEnv(rank=0, local_rank=0, world_size=1)
+    m = {"a": 1.5, "b": -2.0}
+    assert env.average_metrics(m) == m
+    env.barrier()  # no-op without a process group
