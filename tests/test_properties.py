@@ -150,3 +150,37 @@ def test_paired_augment_consistency(im, seed):
     assert any(
         a.shape == v.shape and np.array_equal(a, v) for v in variants
     )
+
+
+@settings(max_examples=20, **COMMON)
+@given(npst.arrays(dtype=np.uint8,
+                   shape=st.tuples(st.integers(2, 24), st.integers(2, 24),
+                                   st.just(3)),
+                   elements=st.integers(100, 103)))
+def test_transform_near_grayscale_path(im):
+    """Near-constant images exercise the reference's grayscale WB branch
+    (saturation quantiles in [0.001, 0.005], data.py:23-44) and the CLAHE
+    uniform-histogram case; outputs must stay valid uint8 everywhere."""
+    wb, gc, he = transform(im)
+    for out in (wb, gc, he):
+        assert out.dtype == np.uint8 and out.shape == im.shape
+
+
+@settings(max_examples=20, **COMMON)
+@given(st.integers(1, 30), st.integers(2, 2**31 - 1))
+def test_synthetic_dataset_deterministic(n, seed):
+    """SyntheticUIEBDataset is a FIXED pseudo-dataset: same (seed, idx) ->
+    identical images across instances, raw_mode yields uint8 HWC pairs."""
+    from waternet_amd.data.dataset import SyntheticUIEBDataset
+
+    a = SyntheticUIEBDataset(n_images=n, im_height=16, im_width=16,
+                             seed=seed, raw_mode=True)
+    b = SyntheticUIEBDataset(n_images=n, im_height=16, im_width=16,
+                             seed=seed, raw_mode=True)
+    idx = n - 1
+    ia, ib = a[idx], b[idx]
+    assert ia["raw"].dtype == torch.uint8
+    assert ia["raw"].shape == (16, 16, 3)
+    assert torch.equal(ia["raw"], ib["raw"])
+    assert torch.equal(ia["ref"], ib["ref"])
+    assert not torch.equal(ia["raw"], ia["ref"])
