@@ -253,3 +253,25 @@ def test_video_pipeline_with_fake_ffmpeg(tmp_path):
     got = np.frombuffer(produced, dtype=np.uint8).reshape(frames, h, w, 3)
     # frames passed through the actual model: right shape, not a copy
     assert not np.array_equal(got, raw)
+
+
+def test_quickstart_example_runs():
+    """examples/quickstart.py (the README/Colab-analog workflow:
+    train -> score -> enhance on synthetic data) completes on CPU."""
+    out = subprocess.run(
+        [sys.executable, str(REPO / "examples" / "quickstart.py")],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "enhanced frame: (64, 64, 3) uint8" in out.stdout
+
+
+def test_hub_demo_example_runs():
+    """examples/hub_demo.py (the reference README torch.hub quickstart
+    analog) completes on CPU."""
+    out = subprocess.run(
+        [sys.executable, str(REPO / "examples" / "hub_demo.py")],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "-> out (112, 112, 3) uint8" in out.stdout
