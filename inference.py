@@ -24,8 +24,10 @@ from waternet_amd.data.bridge import arr2ten, ten2arr
 from waternet_amd.data.transforms import transform
 from waternet_amd.models.waternet import WaterNet
 
-IM_SUFFIXES = [".jpg", ".jpeg", ".png", ".tiff", ".bmp", ".webp"]
-VID_SUFFIXES = [".mp4", ".avi", ".mov", ".mkv", ".webm"]
+# strict supersets of the reference's lists (inference.py:17-18: bmp/jpg/
+# jpeg/png/gif and mp4/mpeg/avi) — PIL takes the first frame of a .gif
+IM_SUFFIXES = [".bmp", ".jpg", ".jpeg", ".png", ".gif", ".tiff", ".webp"]
+VID_SUFFIXES = [".mp4", ".mpeg", ".avi", ".mov", ".mkv", ".webm"]
 
 
 def parse_args(argv=None):
