@@ -19,6 +19,10 @@ from waternet_amd.models.waternet import WaterNet
 def parse_args(argv=None):
     parser = argparse.ArgumentParser()
     parser.add_argument("--weights", type=str, required=True)
+    # accepted-but-unused, as the reference (score.py:99-101: scoring takes
+    # no epochs; the flag exists there and must keep parsing here)
+    parser.add_argument("--epochs", type=int, default=400,
+                        help="(unused; reference CLI compatibility)")
     parser.add_argument("--batch-size", type=int, default=16)
     parser.add_argument("--height", type=int, default=112)
     parser.add_argument("--width", type=int, default=112)

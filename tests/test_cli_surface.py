@@ -275,3 +275,30 @@ def test_hub_demo_example_runs():
     )
     assert out.returncode == 0, out.stderr
     assert "-> out (112, 112, 3) uint8" in out.stdout
+
+
+def test_reference_flag_sets_parse():
+    """Every flag combination the reference CLIs accept must parse here
+    (CLI contract): train.py:164-194, score.py:92-122 (incl. the vestigial
+    --epochs), inference.py:57-80."""
+    sys.path.insert(0, str(REPO))
+    try:
+        import inference as inf_cli
+        import score as score_cli
+        import train as train_cli
+    finally:
+        sys.path.pop(0)
+
+    t = train_cli.parse_args(["--epochs", "400", "--batch-size", "16",
+                              "--height", "112", "--width", "112",
+                              "--weights", "w.pt", "--seed", "5"])
+    assert t.epochs == 400 and t.seed == 5
+
+    s = score_cli.parse_args(["--weights", "w.pt", "--epochs", "400",
+                              "--batch-size", "16", "--height", "112",
+                              "--width", "112", "--seed", "5"])
+    assert s.weights == "w.pt"
+
+    i = inf_cli.parse_args(["--source", "x.png", "--weights", "w.pt",
+                            "--name", "run", "--show-split"])
+    assert i.show_split and i.name == "run"
