@@ -68,3 +68,49 @@ def test_average_metrics_world1_passthrough():
     m = {"a": 1.5, "b": -2.0}
     assert env.average_metrics(m) == m
     env.barrier()  # no-op without a process group
+
+
+def test_convspec_pack_protocol(monkeypatch):
+    """ConvSpec's dirty/version protocol (round-1 advisor subject): packs
+    are cached by (weight._version, data_ptr), refreshed on mutation or
+    mark_dirty, and the buffers are REWRITTEN IN PLACE — the batched
+    pack_all descriptor bakes their device pointers, so the objects must
+    never be reallocated."""
+    import waternet_amd.ops as ops
+    from waternet_amd.ops.conv import ACT_RELU, ConvSpec
+
+    calls = {"fwd": 0, "dgrad": 0}
+
+    class FakeExt:
+        @staticmethod
+        def pack_weight_fwd(w, kp, cp):
+            calls["fwd"] += 1
+            return torch.full((kp * 9 * cp,), float(calls["fwd"]))
+
+        @staticmethod
+        def pack_weight_dgrad(w, kp, cp):
+            calls["dgrad"] += 1
+            return torch.full((cp * 9 * kp,), float(calls["dgrad"]))
+
+    monkeypatch.setattr(ops, "_ext", FakeExt())
+    monkeypatch.setattr(ops, "_tried", True)
+    mod = torch.nn.Conv2d(3, 5, 3, padding="same")
+    spec = ConvSpec(mod, ACT_RELU)
+    assert (spec.Kp, spec.Cp) == (16, 16)  # pow2 >= 16 channel padding
+
+    wp1 = spec.packed_fwd()
+    ptr = wp1.data_ptr()
+    assert calls["fwd"] == 1
+    spec.packed_fwd()
+    assert calls["fwd"] == 1  # cached: same version -> no repack
+
+    with torch.no_grad():
+        mod.weight += 1.0  # bumps weight._version
+    wp2 = spec.packed_fwd()
+    assert calls["fwd"] == 2  # mutation detected
+    assert wp2.data_ptr() == ptr  # rewritten IN PLACE, never reallocated
+
+    spec.mark_dirty()  # FusedAdam's in-place master update (no version bump)
+    spec.packed_fwd()
+    assert calls["fwd"] == 3
+    assert spec.packed_fwd().data_ptr() == ptr
