@@ -3,6 +3,7 @@ composite, the bare (unbatched) tensor bridges, and the torchrun env
 plumbing of the parallel layer."""
 
 import numpy as np
+import pytest
 import torch
 
 from waternet_amd.data.bridge import arr2ten, ten2arr
@@ -159,3 +160,40 @@ def test_pack_descriptor_protocol(monkeypatch):
     for s in specs:
         s.refresh_if_needed()  # must SKIP: versions were recorded
     assert calls["n"] == n_before
+
+
+def test_native_dispatch_policy(monkeypatch):
+    """_use_native: never on CPU; on GPU it REFUSES to silently fall back
+    to eager when the extension is missing (the loud-failure policy the
+    round-end native-code check depends on); WATERNET_AMD_EAGER=1 is the
+    only escape."""
+    import waternet_amd.ops as ops
+    from waternet_amd.models.waternet import _use_native
+
+    class FakeCudaTensor:
+        is_cuda = True
+
+    assert _use_native(torch.zeros(1)) is False  # CPU tensor -> eager
+
+    monkeypatch.setattr(ops, "_ext", None)
+    monkeypatch.setattr(ops, "_tried", True)
+    monkeypatch.setattr(ops, "_load_err", "ImportError('no _C')")
+    with pytest.raises(RuntimeError, match="refusing to fall back"):
+        _use_native(FakeCudaTensor())
+    monkeypatch.setenv("WATERNET_AMD_EAGER", "1")
+    assert _use_native(FakeCudaTensor()) is False  # explicit escape only
+
+
+def test_fused_adam_sync_lr_buffer():
+    """sync_lr mirrors param_groups lr into the device-resident buffer the
+    graphed k_adam reads — only when it actually changed."""
+    from waternet_amd.ops.adam import FusedAdam
+
+    m = torch.nn.Linear(4, 4)
+    o = FusedAdam(m.parameters(), lr=1e-3)
+    assert abs(float(o.lr_buf.item()) - 1e-3) < 1e-9
+    o.sync_lr()  # unchanged: no-op
+    assert abs(float(o.lr_buf.item()) - 1e-3) < 1e-9
+    o.param_groups[0]["lr"] = 2e-4  # what StepLR does on the host
+    o.sync_lr()
+    assert abs(float(o.lr_buf.item()) - 2e-4) < 1e-12
