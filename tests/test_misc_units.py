@@ -196,4 +196,4 @@ def test_fused_adam_sync_lr_buffer():
     assert abs(float(o.lr_buf.item()) - 1e-3) < 1e-9
     o.param_groups[0]["lr"] = 2e-4  # what StepLR does on the host
     o.sync_lr()
-    assert abs(float(o.lr_buf.item()) - 2e-4) < 1e-12
+    assert abs(float(o.lr_buf.item()) - 2e-4) < 1e-9
