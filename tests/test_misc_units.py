@@ -197,3 +197,33 @@ def test_fused_adam_sync_lr_buffer():
     o.param_groups[0]["lr"] = 2e-4  # what StepLR does on the host
     o.sync_lr()
     assert abs(float(o.lr_buf.item()) - 2e-4) < 1e-9
+
+
+def test_native_state_plans():
+    """The native execution plans mirror the module trees exactly:
+    8 CMG specs (ReLU x7 + Sigmoid), 3x3 refiner specs (ReLU), and the
+    VGG plan = 16 convs + 4 pools with ReLUs folded into epilogues."""
+    from waternet_amd.engine.native import WaterNetNativeState, VggNativeState
+    from waternet_amd.models.vgg import PerceptualModel
+    from waternet_amd.models.waternet import WaterNet
+    from waternet_amd.ops.conv import ACT_RELU, ACT_SIGMOID
+
+    st = WaterNetNativeState(WaterNet())
+    assert len(st.cmg_specs) == 8
+    assert [s.act for s in st.cmg_specs] == [ACT_RELU] * 7 + [ACT_SIGMOID]
+    assert [s.ks for s in st.cmg_specs] == [7, 5, 3, 1, 7, 5, 3, 3]
+    assert [s.K for s in st.cmg_specs] == [128, 128, 128, 64, 64, 64, 64, 3]
+    assert set(st.refiner_specs) == {"wb_refiner", "ce_refiner",
+                                     "gc_refiner"}
+    for specs in st.refiner_specs.values():
+        assert [s.ks for s in specs] == [7, 5, 3]
+        assert [(s.C, s.K) for s in specs] == [(6, 32), (32, 32), (32, 3)]
+        assert all(s.act == ACT_RELU for s in specs)
+
+    vst = VggNativeState(PerceptualModel())
+    kinds = [k for k, _ in vst.plan]
+    assert kinds.count("conv") == 16 and kinds.count("pool") == 4
+    # VGG19-E channel ladder, final pool dropped
+    chans = [spec.K for k, spec in vst.plan if k == "conv"]
+    assert chans == [64, 64, 128, 128, 256, 256, 256, 256,
+                     512, 512, 512, 512, 512, 512, 512, 512]
