@@ -302,3 +302,25 @@ def test_reference_flag_sets_parse():
     i = inf_cli.parse_args(["--source", "x.png", "--weights", "w.pt",
                             "--name", "run", "--show-split"])
     assert i.show_split and i.name == "run"
+
+
+def test_engine_fast_on_cpu_clear_error():
+    """--engine fast without a GPU exits with the requirement message
+    instead of failing deep in the engine."""
+    out = subprocess.run(
+        [sys.executable, str(REPO / "train.py"), "--engine", "fast",
+         "--synthetic", "4", "--epochs", "1",
+         "--batch-size", "2", "--height", "32", "--width", "32"],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "--engine fast requires a ROCm GPU" in out.stderr
+
+    out = subprocess.run(
+        [sys.executable, str(REPO / "score.py"), "--engine", "fast",
+         "--weights", "w.pt", "--synthetic", "4",
+         "--batch-size", "2", "--height", "32", "--width", "32"],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "requires a ROCm GPU" in out.stderr
