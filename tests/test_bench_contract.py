@@ -43,3 +43,17 @@ def test_recorded_bench_line_schema():
     # whole-job aggregate sanity: value == total images / elapsed
     assert abs(rec["value"] * rec["ms_per_step"] / 1000.0
                - rec["n_gpus"] * 16) < 1e-6
+
+
+def test_bench_requires_gpu_clear_exit():
+    """On a CPU-only box bench.py exits with a clear message (the driver
+    only runs it on GPU boxes; anything else must not half-run)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "1"],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode != 0
+    assert "requires a ROCm GPU" in out.stderr
