@@ -176,3 +176,31 @@ def test_hubconf_pretrained_offline_error():
 
     with pytest.raises(RuntimeError, match="checkpoint=|pretrained=False"):
         hubconf.waternet(pretrained=True, device="cpu")
+
+
+def test_loader_with_workers_and_collation(tmp_path):
+    """Multi-worker loading (the fast engine's default, train.py:222-230):
+    PIL decode in worker processes, uint8 raw_mode batches collate to
+    (B,H,W,3), ragged tail preserved, epoch re-iteration works with
+    persistent workers."""
+    from PIL import Image
+
+    from waternet_amd.data.dataset import UIEBDataset
+
+    rng = np.random.default_rng(5)
+    (tmp_path / "raw-890").mkdir()
+    (tmp_path / "reference-890").mkdir()
+    for i in range(5):
+        for d in ("raw-890", "reference-890"):
+            Image.fromarray(rng.integers(
+                0, 256, size=(40, 40, 3), dtype=np.uint8
+            )).save(tmp_path / d / f"{i}.png")
+    ds = UIEBDataset(tmp_path / "raw-890", tmp_path / "reference-890",
+                     im_height=32, im_width=32, raw_mode=True)
+    loader = torch.utils.data.DataLoader(ds, batch_size=2, num_workers=2,
+                                         persistent_workers=True)
+    for _ in range(2):  # two epochs through persistent workers
+        sizes = [b["raw"].shape for b in loader]
+        assert sizes == [torch.Size([2, 32, 32, 3])] * 2 + \
+            [torch.Size([1, 32, 32, 3])]  # ragged tail
+    del loader
