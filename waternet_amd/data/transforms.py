@@ -47,15 +47,23 @@ def white_balance_transform(im_rgb: np.ndarray) -> np.ndarray:
             ratio = chan_sums.max() / chan_sums  # inf/nan handled below
         sat_lo = 0.005 * ratio
         sat_hi = 0.005 * ratio
-        flat = im_rgb.reshape(h * w, p).T.astype(np.float64)
+        flat = im_rgb.reshape(h * w, p).T
     else:
         h, w = im_rgb.shape
         p = 1
         sat_lo = np.array([0.001])
         sat_hi = np.array([0.005])
-        flat = im_rgb.reshape(1, h * w).astype(np.float64)
+        flat = im_rgb.reshape(1, h * w)
 
-    out = np.empty_like(flat)
+    # uint8 data: quantiles are order statistics over a 256-bin histogram
+    # (O(n) bincount instead of np.quantile's O(n log n) sort) and the
+    # clip+stretch is a 256-entry float64 LUT gather — BIT-identical to
+    # the per-pixel float64 formula (verified against the direct
+    # np.quantile composition by test_white_balance_reference_semantics
+    # and the hypothesis fuzz).
+    n = flat.shape[1]
+    out = np.empty((p, n), dtype=np.float64)
+    vals = np.arange(256, dtype=np.float64)
     for ch in range(p):
         lo_q, hi_q = sat_lo[ch], 1.0 - sat_hi[ch]
         # Degenerate channels (a zero channel sum makes the sat ratio
@@ -64,16 +72,36 @@ def white_balance_transform(im_rgb: np.ndarray) -> np.ndarray:
         # stretch for that channel.
         if not (np.isfinite(lo_q) and 0.0 <= lo_q <= 0.5):
             lo_q, hi_q = 0.0, 1.0
-        lo_v, hi_v = np.quantile(flat[ch], [lo_q, hi_q])
-        clipped = np.clip(flat[ch], lo_v, hi_v)
-        bottom = clipped.min()
-        top = clipped.max()
+        hist = np.bincount(flat[ch], minlength=256)
+        cum = np.cumsum(hist)
+        lo_v = _quantile_from_hist(cum, n, lo_q)
+        hi_v = _quantile_from_hist(cum, n, hi_q)
+        nz = np.nonzero(hist)[0]
+        vmin, vmax = float(nz[0]), float(nz[-1])
+        bottom = min(max(vmin, lo_v), hi_v)  # clip is monotone
+        top = min(max(vmax, lo_v), hi_v)
         scale = 255.0 / (top - bottom) if top > bottom else 0.0
-        out[ch] = (clipped - bottom) * scale
+        lut = (np.clip(vals, lo_v, hi_v) - bottom) * scale
+        out[ch] = lut[flat[ch]]
 
     if im_rgb.ndim == 3:
         return out.T.reshape(h, w, p).astype(np.uint8)
     return out.reshape(h, w).astype(np.uint8)
+
+
+def _quantile_from_hist(cum: np.ndarray, n: int, q: float) -> float:
+    """np.quantile('linear') over uint8 data from its cumulative histogram:
+    pos = q*(n-1); order statistics a[floor]/a[ceil] via cumulative counts;
+    numpy's _lerp semantics (the t>=0.5 branch) replicated exactly."""
+    pos = q * (n - 1)
+    lo_i = int(np.floor(pos))
+    frac = pos - lo_i
+    hi_i = lo_i + 1 if frac > 0.0 else lo_i
+    v_lo = float(np.searchsorted(cum, lo_i, side="right"))
+    v_hi = float(np.searchsorted(cum, hi_i, side="right"))
+    if frac >= 0.5:  # numpy _lerp's accuracy branch
+        return v_hi - (v_hi - v_lo) * (1.0 - frac)
+    return v_lo + (v_hi - v_lo) * frac
 
 
 # --------------------------------------------------------------------------
@@ -83,10 +111,20 @@ def white_balance_transform(im_rgb: np.ndarray) -> np.ndarray:
 GAMMA = 0.7
 
 
+_GAMMA_LUT = None
+
+
 def gamma_correction(im: np.ndarray) -> np.ndarray:
-    """gc = clip(255 * (im/255)^0.7, 0, 255) truncated to uint8."""
-    gc = np.power(im.astype(np.float64) / 255.0, GAMMA)
-    return np.clip(255.0 * gc, 0, 255).astype(np.uint8)
+    """gc = clip(255 * (im/255)^0.7, 0, 255) truncated to uint8.
+
+    uint8 input -> the map has only 256 possible values: computed once in
+    float64 (bit-identical to the per-pixel formula) and applied as a LUT
+    gather. The GPU kernel uses this same host-computed table."""
+    global _GAMMA_LUT
+    if _GAMMA_LUT is None:
+        v = np.power(np.arange(256, dtype=np.float64) / 255.0, GAMMA)
+        _GAMMA_LUT = np.clip(255.0 * v, 0, 255).astype(np.uint8)
+    return _GAMMA_LUT[im]
 
 
 # --------------------------------------------------------------------------
@@ -128,11 +166,19 @@ def _lab_finv(ft: np.ndarray) -> np.ndarray:
     return np.where(ft > d, ft**3, 3 * d * d * (ft - 4.0 / 29.0))
 
 
+_SRGB_LIN_LUT = None
+
+
 def rgb2lab_u8(rgb: np.ndarray) -> np.ndarray:
     """HWC uint8 RGB -> HWC uint8 LAB with OpenCV 8-bit scaling
-    (L*255/100, a+128, b+128)."""
-    s = rgb.astype(np.float64) / 255.0
-    lin = _srgb_linearize(s)
+    (L*255/100, a+128, b+128). The sRGB linearization of a uint8 input
+    has 256 possible values — applied as a float64 LUT gather,
+    bit-identical to calling _srgb_linearize per pixel."""
+    global _SRGB_LIN_LUT
+    if _SRGB_LIN_LUT is None:
+        _SRGB_LIN_LUT = _srgb_linearize(
+            np.arange(256, dtype=np.float64) / 255.0)
+    lin = _SRGB_LIN_LUT[rgb]
     xyz = lin @ _RGB2XYZ.T
     fxyz = _lab_f(xyz / _WHITE)
     L = 116.0 * fxyz[..., 1] - 16.0
