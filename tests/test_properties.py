@@ -242,3 +242,20 @@ def test_lut_paths_bit_equal_naive_formulas(im, kind):
     np.testing.assert_array_equal(
         white_balance_transform(im),
         exp.T.reshape(h, w, p).astype(np.uint8))
+
+
+@settings(max_examples=25, **COMMON)
+@given(npst.arrays(dtype=np.uint8,
+                   shape=st.tuples(st.integers(2, 40), st.integers(2, 40)),
+                   elements=st.integers(0, 255)))
+def test_wb_grayscale_bit_equal_naive(im):
+    """The histogram fast path on the 2-D grayscale branch (fixed
+    [0.001, 0.005] saturations, data.py:23-44) is bit-identical to the
+    direct np.quantile composition."""
+    flat = im.reshape(-1).astype(np.float64)
+    lo_v, hi_v = np.quantile(flat, [0.001, 1.0 - 0.005])
+    clipped = np.clip(flat, lo_v, hi_v)
+    bottom, top = clipped.min(), clipped.max()
+    scale = 255.0 / (top - bottom) if top > bottom else 0.0
+    expected = ((clipped - bottom) * scale).reshape(im.shape).astype(np.uint8)
+    np.testing.assert_array_equal(white_balance_transform(im), expected)
